@@ -1,0 +1,201 @@
+"""Collective data plane: shard math + RCCL/xGMI transfers.
+
+This replaces the reference's entire message stack for table traffic —
+NetInterface/MPI point-to-point sends (include/multiverso/net/mpi_net.h),
+the worker→communicator→server actor chain, and the
+hand-rolled AllreduceEngine (src/net/allreduce_engine.cpp:31-172).
+
+MI355X-native mapping (BASELINE.json north star):
+- whole-table Get  = all-gather of the HBM-resident server shards
+- whole-table Add  = reduce-scatter of worker deltas (the sum over workers
+  lands directly on the owning shard — no per-worker messages, no server
+  mailbox) followed by the updater kernel on the owner
+- row-keyed Get/Add = all-to-all exchange (keys, then values) — irregular
+  p2p traffic is batched per destination, which matches xGMI's 7
+  point-to-point links far better than a ring of small messages
+- aggregate (MV_Aggregate / -ma mode) = one all-reduce
+
+Sharding parity: contiguous partition with ``total // n`` per server and
+the remainder on the last server — identical to ArrayTable
+(src/table/array_table.cpp:11-21) and MatrixTable row sharding
+(src/table/matrix_table.cpp:24-45), so checkpoint bytes line up.
+
+All collectives can run with ``async_op=True``; on GPU they are enqueued
+through RCCL's stream and a returned handle carries stream-level
+dependencies, so Add/Get overlap with compute (the reference's
+ASyncBuffer/pipeline capability).
+"""
+
+from __future__ import annotations
+
+from typing import Callable, List, Optional, Sequence, Tuple
+
+import torch
+import torch.distributed as dist
+
+
+class ShardSpec:
+    """Contiguous sharding of ``total`` units across ``n`` servers."""
+
+    def __init__(self, total: int, n: int) -> None:
+        self.total = total
+        self.n = n
+        base = total // n
+        self.offsets: List[int] = [r * base for r in range(n)]
+        self.counts: List[int] = [base] * n
+        self.counts[-1] = total - (n - 1) * base
+        self.even = (total % n == 0)
+        self.max_count = max(self.counts)
+
+    def range_of(self, rank: int) -> Tuple[int, int]:
+        return self.offsets[rank], self.counts[rank]
+
+    def owner_of(self, idx: int) -> int:
+        """Owning server of unit ``idx`` (reference matrix_table.cpp:276:
+        dst = key / num_row_each, clamped to the last server)."""
+        base = self.total // self.n
+        return min(idx // base, self.n - 1) if base else self.n - 1
+
+
+class Handle:
+    """Completion handle for an async table op (the reference's Waiter,
+    include/multiverso/util/waiter.h). ``wait()`` resolves the collective
+    and runs any deferred epilogue (e.g. the server-side updater)."""
+
+    def __init__(self, work=None, epilogue: Optional[Callable[[], None]] = None):
+        self._work = work
+        self._epilogue = epilogue
+        self._done = False
+
+    def wait(self) -> None:
+        if self._done:
+            return
+        if self._work is not None:
+            self._work.wait()
+        if self._epilogue is not None:
+            self._epilogue()
+        self._done = True
+
+
+_NOOP = Handle()
+_NOOP._done = True
+
+
+def _dist() -> bool:
+    return dist.is_initialized() and dist.get_world_size() > 1
+
+
+def allgather_shards(out_flat: torch.Tensor, shard_flat: torch.Tensor,
+                     spec: ShardSpec, unit: int, async_op: bool = False) -> Handle:
+    """Gather every server's shard into ``out_flat`` (size total*unit).
+
+    ``unit`` = elements per shard unit (row width for matrices, 1 for
+    arrays). Fast path (even shards): one all_gather_into_tensor straight
+    into the user buffer — zero staging copies. Uneven: padded gather.
+    """
+    if not _dist():
+        off, cnt = spec.range_of(0)
+        out_flat[off * unit:(off + cnt) * unit].copy_(shard_flat)
+        return _NOOP
+    if spec.even and out_flat.is_contiguous():
+        work = dist.all_gather_into_tensor(out_flat, shard_flat, async_op=async_op)
+        return Handle(work) if async_op else _NOOP
+    # uneven: pad each shard to max_count units
+    n = spec.n
+    pad = spec.max_count * unit
+    buf = torch.empty(n * pad, dtype=shard_flat.dtype, device=shard_flat.device)
+    me = torch.zeros(pad, dtype=shard_flat.dtype, device=shard_flat.device)
+    me[:shard_flat.numel()].copy_(shard_flat)
+    work = dist.all_gather_into_tensor(buf, me, async_op=async_op)
+
+    def epilogue() -> None:
+        for r in range(n):
+            off, cnt = spec.range_of(r)
+            out_flat[off * unit:(off + cnt) * unit].copy_(buf[r * pad:r * pad + cnt * unit])
+
+    if async_op:
+        return Handle(work, epilogue)
+    epilogue()
+    return _NOOP
+
+
+def reduce_scatter_delta(delta_flat: torch.Tensor, spec: ShardSpec, unit: int,
+                         async_op: bool = False) -> Tuple[torch.Tensor, Handle]:
+    """Sum ``delta_flat`` across workers and return this rank's shard chunk.
+
+    The whole-table Add path: replaces per-server Request_Add messages
+    (worker.cpp:65-74 → server.cpp:48) with one reduce-scatter over xGMI.
+    """
+    if not _dist():
+        return delta_flat, _NOOP
+    rank = dist.get_rank()
+    if spec.even:
+        off, cnt = spec.range_of(rank)
+        out = torch.empty(cnt * unit, dtype=delta_flat.dtype, device=delta_flat.device)
+        work = dist.reduce_scatter_tensor(out, delta_flat, op=dist.ReduceOp.SUM,
+                                          async_op=async_op)
+        return out, (Handle(work) if async_op else _NOOP)
+    # uneven: pad to n * max_count
+    n, pad = spec.n, spec.max_count * unit
+    buf = torch.zeros(n * pad, dtype=delta_flat.dtype, device=delta_flat.device)
+    for r in range(n):
+        off, cnt = spec.range_of(r)
+        buf[r * pad:r * pad + cnt * unit].copy_(delta_flat[off * unit:(off + cnt) * unit])
+    off, cnt = spec.range_of(rank)
+    out = torch.empty(pad, dtype=delta_flat.dtype, device=delta_flat.device)
+    work = dist.reduce_scatter_tensor(out, buf, op=dist.ReduceOp.SUM, async_op=async_op)
+    out = out[:cnt * unit]
+    return out, (Handle(work) if async_op else _NOOP)
+
+
+def all_to_all_rows(row_ids: torch.Tensor, values: Optional[torch.Tensor],
+                    spec: ShardSpec, unit: int
+                    ) -> Tuple[torch.Tensor, Optional[torch.Tensor],
+                               List[int], torch.Tensor]:
+    """Exchange keyed rows with their owners.
+
+    Returns (incoming_ids, incoming_values, out_split_sizes, send_order):
+    every rank receives the (ids, values) destined for its shard. Used for
+    row-subset Add (values != None) and the request half of row-subset Get.
+    ``send_order`` is the permutation that grouped our row_ids by owner (so
+    a Get reply can be scattered back to the caller's order).
+    """
+    device = row_ids.device
+    if not _dist():
+        order = torch.argsort(row_ids, stable=True)
+        return row_ids, values, [row_ids.numel()], torch.arange(
+            row_ids.numel(), device=device)
+    n = dist.get_world_size()
+    owners = torch.div(row_ids, max(spec.total // spec.n, 1),
+                       rounding_mode="floor").clamp_(max=spec.n - 1)
+    order = torch.argsort(owners, stable=True)
+    sorted_ids = row_ids[order]
+    send_counts = torch.bincount(owners, minlength=n)
+    send_sizes = send_counts.tolist()
+    # exchange counts
+    recv_counts = torch.empty(n, dtype=send_counts.dtype, device=device)
+    dist.all_to_all_single(recv_counts, send_counts.to(device))
+    recv_sizes = recv_counts.tolist()
+    in_ids = torch.empty(sum(recv_sizes), dtype=row_ids.dtype, device=device)
+    dist.all_to_all_single(in_ids, sorted_ids, recv_sizes, send_sizes)
+    in_vals = None
+    if values is not None:
+        sorted_vals = values.reshape(row_ids.numel(), unit)[order].reshape(-1)
+        in_vals = torch.empty(sum(recv_sizes) * unit, dtype=values.dtype, device=device)
+        dist.all_to_all_single(in_vals, sorted_vals.contiguous(),
+                               [c * unit for c in recv_sizes],
+                               [c * unit for c in send_sizes])
+    return in_ids, in_vals, recv_sizes, order
+
+
+def all_to_all_values(values: torch.Tensor, send_sizes: List[int],
+                      recv_sizes: List[int], unit: int) -> torch.Tensor:
+    """Reply half of a keyed Get: send gathered rows back to requesters."""
+    if not _dist():
+        return values
+    out = torch.empty(sum(recv_sizes) * unit, dtype=values.dtype,
+                      device=values.device)
+    dist.all_to_all_single(out, values.contiguous(),
+                           [c * unit for c in recv_sizes],
+                           [c * unit for c in send_sizes])
+    return out

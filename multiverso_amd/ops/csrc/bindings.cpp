@@ -1,0 +1,116 @@
+// Torch binding layer for the gfx950 HIP kernels (kernels.hip).
+//
+// This translation unit is compiled by torch.utils.cpp_extension (it needs
+// the ATen headers); the kernels themselves are hipcc-compiled pure HIP in
+// kernels.hip and linked in as an object, so no hipify/CUDA-compat layer
+// ever touches the device code.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <cstdint>
+
+extern "C" {
+void mv_launch_add(float*, const float*, int64_t, hipStream_t);
+void mv_launch_sgd(float*, const float*, int64_t, hipStream_t);
+void mv_launch_momentum(float*, float*, const float*, float, int64_t, hipStream_t);
+void mv_launch_adagrad(float*, float*, const float*, float, float, float,
+                       int64_t, hipStream_t);
+void mv_launch_row_gather(float*, const float*, const int64_t*, int64_t,
+                          int64_t, hipStream_t);
+void mv_launch_row_scatter_add(float*, const float*, const int64_t*, float,
+                               int64_t, int64_t, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+}
+
+void add_inplace(torch::Tensor data, torch::Tensor delta) {
+  check_f32(data, "data"); check_f32(delta, "delta");
+  TORCH_CHECK(data.numel() == delta.numel(), "size mismatch");
+  mv_launch_add(data.data_ptr<float>(), delta.data_ptr<float>(),
+                data.numel(), cur_stream());
+}
+
+void sgd_update(torch::Tensor data, torch::Tensor delta) {
+  check_f32(data, "data"); check_f32(delta, "delta");
+  TORCH_CHECK(data.numel() == delta.numel(), "size mismatch");
+  mv_launch_sgd(data.data_ptr<float>(), delta.data_ptr<float>(),
+                data.numel(), cur_stream());
+}
+
+void momentum_update(torch::Tensor data, torch::Tensor m, torch::Tensor delta,
+                     double mu) {
+  check_f32(data, "data"); check_f32(m, "m"); check_f32(delta, "delta");
+  TORCH_CHECK(data.numel() == delta.numel() && data.numel() == m.numel(),
+              "size mismatch");
+  mv_launch_momentum(data.data_ptr<float>(), m.data_ptr<float>(),
+                     delta.data_ptr<float>(), (float)mu, data.numel(),
+                     cur_stream());
+}
+
+void adagrad_update(torch::Tensor data, torch::Tensor gsq, torch::Tensor delta,
+                    double lr, double rho, double eps) {
+  check_f32(data, "data"); check_f32(gsq, "gsq"); check_f32(delta, "delta");
+  TORCH_CHECK(data.numel() == delta.numel() && data.numel() == gsq.numel(),
+              "size mismatch");
+  mv_launch_adagrad(data.data_ptr<float>(), gsq.data_ptr<float>(),
+                    delta.data_ptr<float>(), (float)lr, (float)rho, (float)eps,
+                    data.numel(), cur_stream());
+}
+
+torch::Tensor row_gather(torch::Tensor shard, torch::Tensor rows) {
+  check_f32(shard, "shard");
+  TORCH_CHECK(shard.dim() == 2, "shard must be 2-D");
+  TORCH_CHECK(rows.scalar_type() == torch::kInt64 && rows.is_cuda() &&
+              rows.is_contiguous(), "rows must be contiguous int64 on GPU");
+  auto out = torch::empty({rows.numel(), shard.size(1)}, shard.options());
+  mv_launch_row_gather(out.data_ptr<float>(), shard.data_ptr<float>(),
+                       rows.data_ptr<int64_t>(), rows.numel(), shard.size(1),
+                       cur_stream());
+  return out;
+}
+
+void row_gather_out(torch::Tensor out, torch::Tensor shard, torch::Tensor rows) {
+  check_f32(shard, "shard"); check_f32(out, "out");
+  TORCH_CHECK(shard.dim() == 2, "shard must be 2-D");
+  TORCH_CHECK(rows.scalar_type() == torch::kInt64 && rows.is_cuda() &&
+              rows.is_contiguous(), "rows must be contiguous int64 on GPU");
+  TORCH_CHECK(out.numel() == rows.numel() * shard.size(1), "out size mismatch");
+  mv_launch_row_gather(out.data_ptr<float>(), shard.data_ptr<float>(),
+                       rows.data_ptr<int64_t>(), rows.numel(), shard.size(1),
+                       cur_stream());
+}
+
+void row_scatter_add(torch::Tensor shard, torch::Tensor rows,
+                     torch::Tensor vals, double sign) {
+  check_f32(shard, "shard"); check_f32(vals, "vals");
+  TORCH_CHECK(shard.dim() == 2, "shard must be 2-D");
+  TORCH_CHECK(rows.scalar_type() == torch::kInt64 && rows.is_cuda() &&
+              rows.is_contiguous(), "rows must be contiguous int64 on GPU");
+  TORCH_CHECK(vals.numel() == rows.numel() * shard.size(1), "vals size mismatch");
+  mv_launch_row_scatter_add(shard.data_ptr<float>(), vals.data_ptr<float>(),
+                            rows.data_ptr<int64_t>(), (float)sign,
+                            rows.numel(), shard.size(1), cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("add_inplace", &add_inplace, "K1: data += delta (fp32, fused)");
+  m.def("sgd_update", &sgd_update, "K2: data -= delta");
+  m.def("momentum_update", &momentum_update, "K3: fused momentum update");
+  m.def("adagrad_update", &adagrad_update, "K4: fused adagrad update");
+  m.def("row_gather", &row_gather, "K6: out[i] = shard[rows[i]]");
+  m.def("row_gather_out", &row_gather_out, "K6 (preallocated out)");
+  m.def("row_scatter_add", &row_scatter_add,
+        "K5: shard[rows[i]] += sign*vals[i] (atomic)");
+}

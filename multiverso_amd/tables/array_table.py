@@ -1,0 +1,97 @@
+"""ArrayTable — 1-D dense table, contiguous-sharded across ranks.
+
+Capability parity with the reference ArrayTable
+(src/table/array_table.cpp, include/multiverso/table/array_table.h):
+whole-table Get/Add only (the reference always sends key −1), contiguous
+partition ``size/num_servers`` with the remainder on the last server
+(array_table.cpp:11-21), server-side updater on Add (:116-127), Access
+copy-out on Get (:130-141), raw-bytes Store/Load (:144-151).
+
+MI355X mapping: Get = all-gather of shards into the caller's buffer;
+Add = reduce-scatter of the delta followed by one fused updater kernel on
+the owned shard. Both can be issued async and overlap with compute.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from ..comm import Handle, ShardSpec, allgather_shards, reduce_scatter_delta
+from ..dashboard import monitor
+from ..log import CHECK
+from ..updaters import AddOption
+from .base import Table
+
+
+class ArrayTable(Table):
+    def __init__(self, size: int, dtype: torch.dtype = torch.float32,
+                 updater_type: Optional[str] = None) -> None:
+        super().__init__(updater_type)
+        CHECK(size >= self.zoo.num_servers,
+              f"table size {size} must be >= num_servers "
+              f"{self.zoo.num_servers} (reference array_table.cpp:14)")
+        self.size = size
+        self.dtype = dtype
+        self.spec = ShardSpec(size, self.zoo.num_servers)
+        off, cnt = self.spec.range_of(self.zoo.server_id)
+        self.shard = torch.zeros(cnt, dtype=dtype, device=self.device)
+        self._make_updater(self.shard)
+
+    # ---- worker ops ----
+    def get(self, out: Optional[torch.Tensor] = None,
+            async_op: bool = False):
+        """Whole-table Get (array_table.cpp:24-66 semantics)."""
+        self.flush()
+        if out is None:
+            out = torch.empty(self.size, dtype=self.dtype, device=self.device)
+        CHECK(out.numel() == self.size, "Get buffer size mismatch")
+        with monitor("worker.get"):
+            h = allgather_shards(out.view(-1), self.shard, self.spec, 1,
+                                 async_op=async_op)
+        if async_op:
+            self._track(h)
+            return out, h
+        return out
+
+    def add(self, delta: torch.Tensor, option: Optional[AddOption] = None,
+            async_op: bool = False) -> Handle:
+        """Whole-table Add: reduce-scatter + updater (server.cpp:48 →
+        array_table.cpp:116-127)."""
+        CHECK(delta.numel() == self.size, "Add delta size mismatch")
+        delta = delta.to(self.device, self.dtype).contiguous().view(-1)
+        with monitor("worker.add"):
+            chunk, h = reduce_scatter_delta(delta, self.spec, 1,
+                                            async_op=async_op)
+            if async_op:
+                upd, opt = self.updater, option
+
+                def epilogue() -> None:
+                    with monitor("server.update"):
+                        upd.update(chunk, opt)
+
+                # inner Handle resolves the collective; epilogue applies the
+                # updater exactly once (Handle latches on _done).
+                return self._track(Handle(h, epilogue))
+            with monitor("server.update"):
+                self.updater.update(chunk, option)
+            return Handle()
+
+    # ---- checkpoint (Store/Load, array_table.cpp:144-151) ----
+    def store(self, path: str) -> None:
+        """Write whole-table raw bytes (== concatenated server shard dumps,
+        byte-identical to the reference's per-shard Store layout)."""
+        full = self.get()
+        if self.zoo.rank == 0:
+            full.cpu().numpy().tofile(path)
+        self.zoo.barrier()
+
+    def load(self, path: str) -> None:
+        import numpy as np
+        arr = np.fromfile(path, dtype=str(self.dtype).replace("torch.", ""))
+        CHECK(arr.size == self.size, "checkpoint size mismatch")
+        full = torch.from_numpy(arr).to(self.device)
+        off, cnt = self.spec.range_of(self.zoo.server_id)
+        self.shard.copy_(full[off:off + cnt])
+        self.zoo.barrier()

@@ -1,0 +1,150 @@
+"""Server-side updaters — each a single fused CDNA4 HIP kernel on GPU.
+
+Capability parity with the reference updater family (selected by the
+``updater_type`` flag, src/updater/updater.cpp:46-57):
+
+- default: ``data += delta``            (updater.cpp:22-29, K1)
+- sgd:     ``data -= delta``            (sgd_updater.h:14-19, K2 — the
+           worker pre-scales delta by lr)
+- momentum:``m = mu*m + (1-mu)*delta; data -= m``
+           (momentum_updater.h:17-25, K3)
+- adagrad: ``g = delta/lr; G += g*g; data -= rho * g / sqrt(G + eps)``
+           (adagrad_updater.h:23-41, K4). The reference decrements G and
+           copies the accumulator row by value so the state update is lost
+           (SURVEY.md §2.3 flags this as a bug to fix) — we keep the
+           intended accumulate semantics. In the collective data plane the
+           reduce-scattered delta is the sum over workers, so there is one
+           accumulator per shard rather than one per (worker, shard).
+
+``AddOption`` keeps the reference's 20-byte wire envelope
+(include/multiverso/updater/updater.h:10-70) for C-API parity.
+
+Each updater's ``update`` is ONE pass over the shard: on GPU it dispatches
+to the in-tree HIP extension (multiverso_amd.ops) — fused read-modify-write
+kernels, float4-vectorized, grid-stride (memory-bound per the CDNA4 guide:
+the bound is HBM3E bytes, so fusing m/G state updates into the same pass
+halves traffic vs. composing torch ops). CPU fallback uses torch ops and
+is numerically identical in fp32.
+"""
+
+from __future__ import annotations
+
+import struct
+from typing import Dict, Optional, Type
+
+import torch
+
+
+class AddOption:
+    """20-byte (5 x 4B) add-option envelope; field order matches the
+    reference union layout: worker_id(i), momentum(f), lr(f), rho(f),
+    lambda(f)."""
+
+    __slots__ = ("worker_id", "momentum", "learning_rate", "rho", "lambda_")
+
+    def __init__(self, worker_id: int = 0, momentum: float = 0.0,
+                 learning_rate: float = 0.01, rho: float = 0.1,
+                 lambda_: float = 0.1) -> None:
+        self.worker_id = worker_id
+        self.momentum = momentum
+        self.learning_rate = learning_rate
+        self.rho = rho
+        self.lambda_ = lambda_
+
+    def to_bytes(self) -> bytes:
+        return struct.pack("<iffff", self.worker_id, self.momentum,
+                           self.learning_rate, self.rho, self.lambda_)
+
+    @classmethod
+    def from_bytes(cls, b: bytes) -> "AddOption":
+        w, m, lr, rho, lam = struct.unpack("<iffff", b[:20])
+        return cls(w, m, lr, rho, lam)
+
+
+def _hip_ops():
+    """In-tree HIP extension, or None on CPU-only hosts. On a GPU host a
+    missing extension is a hard error (no silent eager fallback)."""
+    from .. import ops
+    return ops.module(required=torch.cuda.is_available())
+
+
+class Updater:
+    name = "default"
+
+    def __init__(self, shard: torch.Tensor) -> None:
+        self.shard = shard
+
+    def update(self, delta: torch.Tensor, option: Optional[AddOption]) -> None:
+        if self.shard.is_cuda:
+            _hip_ops().add_inplace(self.shard, delta)
+        else:
+            self.shard.add_(delta)
+
+    def access(self, out: torch.Tensor) -> None:
+        """K7: shard copy-out (updater.cpp:32-36)."""
+        out.copy_(self.shard)
+
+
+class SGDUpdater(Updater):
+    name = "sgd"
+
+    def update(self, delta: torch.Tensor, option: Optional[AddOption]) -> None:
+        if self.shard.is_cuda:
+            _hip_ops().sgd_update(self.shard, delta)
+        else:
+            self.shard.sub_(delta)
+
+
+class MomentumUpdater(Updater):
+    name = "momentum"
+
+    def __init__(self, shard: torch.Tensor) -> None:
+        super().__init__(shard)
+        self.smooth_gradient = torch.zeros_like(shard)
+
+    def update(self, delta: torch.Tensor, option: Optional[AddOption]) -> None:
+        mu = option.momentum if option else 0.0
+        if self.shard.is_cuda:
+            _hip_ops().momentum_update(self.shard, self.smooth_gradient,
+                                       delta, float(mu))
+        else:
+            self.smooth_gradient.mul_(mu).add_(delta, alpha=1.0 - mu)
+            self.shard.sub_(self.smooth_gradient)
+
+
+class AdaGradUpdater(Updater):
+    name = "adagrad"
+    EPS = 1e-6
+
+    def __init__(self, shard: torch.Tensor) -> None:
+        super().__init__(shard)
+        self.g_sqr = torch.zeros_like(shard)
+
+    def update(self, delta: torch.Tensor, option: Optional[AddOption]) -> None:
+        opt = option or AddOption()
+        lr, rho = opt.learning_rate, opt.rho
+        if self.shard.is_cuda:
+            _hip_ops().adagrad_update(self.shard, self.g_sqr, delta,
+                                      float(lr), float(rho), self.EPS)
+        else:
+            g = delta / lr
+            self.g_sqr.add_(g * g)
+            self.shard.sub_(rho * g / torch.sqrt(self.g_sqr + self.EPS))
+
+
+_REGISTRY: Dict[str, Type[Updater]] = {
+    "default": Updater,
+    "sgd": SGDUpdater,
+    "momentum": MomentumUpdater,
+    "adagrad": AdaGradUpdater,
+}
+
+
+def create_updater(name: str, shard: torch.Tensor) -> Updater:
+    """Factory keyed by the ``updater_type`` flag (updater.cpp:46-57)."""
+    try:
+        cls = _REGISTRY[name]
+    except KeyError:
+        raise ValueError(f"unknown updater_type '{name}' "
+                         f"(have {sorted(_REGISTRY)})") from None
+    return cls(shard)

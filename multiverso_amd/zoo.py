@@ -1,0 +1,175 @@
+"""Zoo — the per-process runtime singleton.
+
+Capability parity with the reference Zoo (src/zoo.cpp:41-188,
+include/multiverso/zoo.h): system bring-up/teardown, rank↔role bookkeeping,
+barrier, table registration, and the model-average aggregate entry point.
+
+MI355X-native redesign: instead of MPI init plus a
+rank-0 controller actor doing Control_Register round-trips, each process is
+one rank of a ``torch.distributed`` process group — RCCL over xGMI on GPU
+nodes (backend "nccl" IS RCCL on ROCm), gloo on CPU. Rendezvous (the
+reference's RegisterNode / RegisterController, zoo.cpp:116-145 /
+controller.cpp:46-72) is the torchrun env rendezvous; Barrier
+(zoo.cpp:164-176) is ``dist.barrier``. Every rank is worker+server
+(ps_role=default → Role::ALL, zoo.cpp:23,29-35); table shards live in that
+rank's HBM. There is no message-passing actor chain on the data path at
+all: Get/Add are collectives issued from the caller's thread onto a side
+HIP stream (comm.py).
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+import threading
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from .configure import get_flag, parse_cmd_flags, set_flag
+from .log import CHECK, log
+
+
+class Role:
+    NONE = 0
+    WORKER = 1
+    SERVER = 2
+    ALL = 3
+
+
+class Zoo:
+    _instance: Optional["Zoo"] = None
+    _lock = threading.Lock()
+
+    def __init__(self) -> None:
+        self.started = False
+        self.rank = 0
+        self.size = 1
+        self.role = Role.ALL
+        self.device: torch.device = torch.device("cpu")
+        self.comm_stream = None  # side HIP stream for collectives
+        self._tables: Dict[int, object] = {}
+        self._next_table_id = 0
+        self._owns_pg = False
+        self.backend = "none"
+
+    # ---- singleton ----
+    @classmethod
+    def get(cls) -> "Zoo":
+        with cls._lock:
+            if cls._instance is None:
+                cls._instance = Zoo()
+            return cls._instance
+
+    # ---- lifecycle ----
+    def start(self, argv: Optional[List[str]] = None,
+              backend: Optional[str] = None) -> List[str]:
+        """MV_Init equivalent (multiverso.cpp:11, zoo.cpp:41-102)."""
+        if self.started:
+            return argv or []
+        rest = parse_cmd_flags(argv or [])
+
+        role = get_flag("ps_role")
+        self.role = {"default": Role.ALL, "worker": Role.WORKER,
+                     "server": Role.SERVER, "none": Role.NONE}.get(role, Role.ALL)
+        CHECK(self.role == Role.ALL,
+              "MI355X rebuild runs every rank as worker+server (role=default); "
+              f"got ps_role={role}")
+
+        world_size = int(os.environ.get("WORLD_SIZE", "1"))
+        cuda = torch.cuda.is_available()
+
+        if cuda:
+            local_rank = int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", "0")))
+            torch.cuda.set_device(local_rank)
+            self.device = torch.device("cuda", local_rank)
+        else:
+            self.device = torch.device("cpu")
+
+        if dist.is_initialized():
+            self._owns_pg = False
+            self.backend = dist.get_backend()
+        elif world_size > 1 or "MASTER_ADDR" in os.environ:
+            self.backend = backend or os.environ.get(
+                "MV_BACKEND", "nccl" if cuda else "gloo")
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29517")
+            dist.init_process_group(
+                backend=self.backend,
+                timeout=datetime.timedelta(seconds=300),
+            )
+            self._owns_pg = True
+        else:
+            self.backend = "none"  # single-process degenerate mode
+
+        if dist.is_initialized():
+            self.rank = dist.get_rank()
+            self.size = dist.get_world_size()
+        else:
+            self.rank, self.size = 0, 1
+
+        if cuda:
+            self.comm_stream = torch.cuda.Stream(device=self.device)
+
+        self.started = True
+        log.debug(f"Zoo started: rank {self.rank}/{self.size} "
+                  f"backend={self.backend} device={self.device}")
+        self.barrier()
+        return rest
+
+    def stop(self, finalize_net: bool = True) -> None:
+        """MV_ShutDown equivalent (zoo.cpp:104-161)."""
+        if not self.started:
+            return
+        self.barrier()
+        from .tables.base import free_tables
+        free_tables()
+        self._tables.clear()
+        self._next_table_id = 0
+        if self._owns_pg and dist.is_initialized() and finalize_net:
+            dist.destroy_process_group()
+        self.started = False
+
+    # ---- bookkeeping (zoo.h:19-85) ----
+    @property
+    def num_workers(self) -> int:
+        return self.size
+
+    @property
+    def num_servers(self) -> int:
+        return self.size
+
+    @property
+    def worker_id(self) -> int:
+        return self.rank
+
+    @property
+    def server_id(self) -> int:
+        return self.rank
+
+    def register_table(self, table) -> int:
+        tid = self._next_table_id
+        self._next_table_id += 1
+        self._tables[tid] = table
+        return tid
+
+    # ---- collectives ----
+    def barrier(self) -> None:
+        if dist.is_initialized():
+            if self.backend == "nccl":
+                dist.barrier(device_ids=[self.device.index])
+                torch.cuda.synchronize(self.device)
+            else:
+                dist.barrier()
+
+    def aggregate(self, tensor: torch.Tensor) -> torch.Tensor:
+        """In-place sum-allreduce — MV_Aggregate (src/multiverso.cpp:53-56,
+        net.cpp:27-41). On GPU this is one rcclAllReduce over xGMI."""
+        if dist.is_initialized():
+            dist.all_reduce(tensor, op=dist.ReduceOp.SUM)
+        return tensor
+
+    @property
+    def sync_mode(self) -> bool:
+        return bool(get_flag("sync"))

@@ -1,0 +1,52 @@
+import os
+import socket
+import sys
+
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an MI355X GPU (run with -m gpu)")
+
+
+def pytest_collection_modifyitems(config, items):
+    import torch
+    if torch.cuda.is_available():
+        return
+    skip = pytest.mark.skip(reason="no GPU in this container")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip)
+
+
+def free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _dist_entry(rank: int, fn, world_size: int, port: int, backend: str):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["MV_BACKEND"] = backend
+    fn(rank, world_size)
+
+
+def run_dist(fn, world_size: int = 2, backend: str = "gloo", timeout: int = 120):
+    """Run ``fn(rank, world_size)`` in world_size fresh processes over gloo.
+
+    This is the rebuild's analog of the reference's `mpirun -np N` CLI
+    tests (SURVEY.md §4 tier 2).
+    """
+    import torch.multiprocessing as mp
+    port = free_port()
+    mp.start_processes(_dist_entry, args=(fn, world_size, port, backend),
+                       nprocs=world_size, join=True, start_method="spawn")

@@ -1,0 +1,143 @@
+"""Multi-process (world_size=2, gloo) integration tests — the rebuild's
+equivalent of the reference's `mpirun -np N` CLI tests (Test/, SURVEY.md §4
+tier 2). Exact-value oracles scaled by world size follow the canonical
+pattern of Test/test_array_table.cpp:31-42 and
+binding/python tests (test_multiverso.py:33)."""
+
+import numpy as np
+import pytest
+import torch
+
+from conftest import run_dist
+
+
+# ---- worker functions (must be top-level for spawn pickling) ----
+
+def _array_sync_oracle(rank, world):
+    import multiverso_amd as mv
+    mv.init(sync=True)
+    t = mv.ArrayTable(10)
+    for it in range(3):
+        t.add(torch.full((10,), 2.0)).wait()
+        got = t.get()
+        expect = torch.full((10,), 2.0 * (it + 1) * world)
+        assert torch.equal(got, expect), (rank, it, got)
+    mv.shutdown()
+
+
+def _array_uneven(rank, world):
+    import multiverso_amd as mv
+    mv.init(sync=True)
+    t = mv.ArrayTable(11)  # uneven: shards 5 + 6
+    t.add(torch.arange(11, dtype=torch.float32))
+    got = t.get()
+    assert torch.equal(got, torch.arange(11, dtype=torch.float32) * world)
+    mv.shutdown()
+
+
+def _matrix_whole_and_rows(rank, world):
+    import multiverso_amd as mv
+    mv.init(sync=True)
+    t = mv.MatrixTable(9, 4)  # uneven rows: 4 + 5
+    delta = torch.ones(9, 4)
+    t.add(delta)
+    got = t.get()
+    assert torch.equal(got, torch.full((9, 4), float(world)))
+    # row ops: each rank adds to different rows
+    rows = [rank, 8 - rank]
+    t.add_rows(rows, torch.full((2, 4), 10.0))
+    got = t.get_rows([0, 1, 8])
+    expect = torch.full((3, 4), float(world))
+    expect[0] += 10.0          # row 0 touched by rank 0
+    expect[1] += 10.0          # row 1 touched by rank 1
+    expect[2] += 10.0          # row 8 touched by rank 0 (8-0)
+    if world == 1:
+        expect = torch.full((3, 4), 1.0)
+        expect[0] += 20.0
+    assert torch.equal(got, expect), (rank, got)
+    mv.shutdown()
+
+
+def _matrix_same_rows(rank, world):
+    import multiverso_amd as mv
+    mv.init(sync=True)
+    t = mv.MatrixTable(6, 2)
+    # every rank adds the same rows -> sums
+    t.add_rows([2, 5], torch.ones(2, 2))
+    got = t.get_rows([2, 5])
+    assert torch.equal(got, torch.full((2, 2), float(world)))
+    mv.shutdown()
+
+
+def _kv(rank, world):
+    import multiverso_amd as mv
+    mv.init()
+    t = mv.KVTable()
+    t.add([1, 2, 3], [1.0, 2.0, 3.0])
+    got = t.get([1, 2, 3])
+    assert got == {1: 1.0 * world, 2: 2.0 * world, 3: 3.0 * world}
+    mv.shutdown()
+
+
+def _aggregate(rank, world):
+    import multiverso_amd as mv
+    mv.init()
+    x = torch.ones(5, dtype=torch.int32)
+    mv.aggregate(x)
+    assert torch.equal(x, torch.full((5,), world, dtype=torch.int32))
+    mv.shutdown()
+
+
+def _handler_init_protocol(rank, world):
+    import multiverso_amd as mv
+    mv.init(sync=True)
+    h = mv.ArrayTableHandler(8, init_value=np.full(8, 3.0, dtype=np.float32))
+    got = h.get()
+    # master added init, others zeros -> value present exactly once
+    assert torch.equal(got, torch.full((8,), 3.0)), (rank, got)
+    mv.shutdown()
+
+
+def _async_pipeline(rank, world):
+    import multiverso_amd as mv
+    mv.init()
+    t = mv.ArrayTable(16)
+    for _ in range(4):
+        t.add(torch.ones(16), async_op=True)
+    got = t.get()  # flushes pending
+    assert torch.equal(got, torch.full((16,), 4.0 * world))
+    mv.shutdown()
+
+
+# ---- tests ----
+
+def test_array_sync_oracle():
+    run_dist(_array_sync_oracle, 2)
+
+
+def test_array_uneven_shards():
+    run_dist(_array_uneven, 2)
+
+
+def test_matrix_whole_and_rows():
+    run_dist(_matrix_whole_and_rows, 2)
+
+
+def test_matrix_same_rows():
+    run_dist(_matrix_same_rows, 2)
+
+
+def test_kv():
+    run_dist(_kv, 2)
+
+
+def test_aggregate():
+    run_dist(_aggregate, 2)
+
+
+def test_handler_init_protocol():
+    run_dist(_handler_init_protocol, 2)
+
+
+def test_async_pipeline():
+    run_dist(_async_pipeline, 2)
