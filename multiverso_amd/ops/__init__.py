@@ -37,6 +37,7 @@ def build(verbose: bool = False) -> str:
 
     os.makedirs(_BUILD, exist_ok=True)
     os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+    rocm_home = os.environ.get("ROCM_HOME", "/opt/rocm")
 
     kernels_src = os.path.join(_CSRC, "kernels.hip")
     kernels_obj = os.path.join(_BUILD, "kernels.hip.o")
@@ -52,8 +53,10 @@ def build(verbose: bool = False) -> str:
     mod = cpp_extension.load(
         name=_MOD_NAME,
         sources=[os.path.join(_CSRC, "bindings.cpp")],
-        extra_objects=[kernels_obj],
-        extra_cflags=["-O3"],
+        extra_cflags=["-O3", "-D__HIP_PLATFORM_AMD__=1"],
+        extra_include_paths=[os.path.join(rocm_home, "include")],
+        extra_ldflags=[kernels_obj,
+                       f"-L{os.path.join(rocm_home, 'lib')}", "-lamdhip64"],
         build_directory=_BUILD,
         verbose=verbose,
     )
