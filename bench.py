@@ -34,6 +34,10 @@ def parse_args():
     p.add_argument("--updater", type=str, default="sgd")
     p.add_argument("--app", type=str, default="matrix",
                    choices=["matrix", "wordembedding"])
+    p.add_argument("--vocab", type=int, default=1_000_000)
+    p.add_argument("--dim", type=int, default=200)
+    p.add_argument("--block-words", dest="block_words", type=int,
+                   default=500_000)
     return p.parse_args()
 
 

@@ -1,0 +1,90 @@
+"""Corpus readers and block streaming.
+
+Capability parity with the reference Reader/DataBlock pipeline
+(Applications/WordEmbedding/src/reader.cpp, data_block.h, the loader
+thread of distributed_wordembedding.cpp:33-56): sentence-block streaming
+with a bounded word budget per block, plus a synthetic Zipf corpus for
+benchmarks (no network → no downloadable datasets; BASELINE.json says
+synthetic data / random-init weights)."""
+
+from __future__ import annotations
+
+from typing import Iterator, List, Optional, Tuple
+
+import torch
+
+from .dictionary import Dictionary
+
+MAX_SENTENCE_LEN = 1000  # reference constant.h kMaxSentenceLength
+
+
+def tokenize_file(path: str) -> Iterator[str]:
+    with open(path) as f:
+        for line in f:
+            yield from line.split()
+
+
+class TextBlockReader:
+    """Streams (words, sent_ids) blocks of ~block_words tokens from a text
+    corpus, dictionary-mapped, unknown words dropped. Each rank reads its
+    own interleaved stripe of sentences (rank r takes sentence s where
+    s % world == r) — the reference's per-rank file offsets
+    (reader.cpp) redesigned for shared storage."""
+
+    def __init__(self, path: str, dictionary: Dictionary, block_words: int,
+                 rank: int = 0, world: int = 1) -> None:
+        self.path = path
+        self.dict = dictionary
+        self.block_words = block_words
+        self.rank = rank
+        self.world = world
+
+    def blocks(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:
+        words: List[int] = []
+        sids: List[int] = []
+        sid = 0
+        taken = 0
+        with open(self.path) as f:
+            for snum, line in enumerate(f):
+                if snum % self.world != self.rank:
+                    continue
+                ids = [self.dict.get_id(w) for w in line.split()]
+                ids = [i for i in ids if i >= 0][:MAX_SENTENCE_LEN]
+                if not ids:
+                    continue
+                words.extend(ids)
+                sids.extend([sid] * len(ids))
+                sid += 1
+                taken += len(ids)
+                if taken >= self.block_words:
+                    yield (torch.tensor(words, dtype=torch.int64),
+                           torch.tensor(sids, dtype=torch.int64))
+                    words, sids, taken = [], [], 0
+        if words:
+            yield (torch.tensor(words, dtype=torch.int64),
+                   torch.tensor(sids, dtype=torch.int64))
+
+
+def synthetic_block(vocab_size: int, n_words: int, sent_len: int = 30,
+                    seed: int = 0,
+                    device: Optional[torch.device] = None
+                    ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Zipf-distributed synthetic corpus block (frequency rank ~ 1/r),
+    matching the shape word2vec sees on natural text."""
+    device = device or torch.device("cpu")
+    g = torch.Generator(device=device).manual_seed(seed)
+    # inverse-CDF Zipf: id = floor(exp(u * log(V))) - 1 approximates 1/r
+    u = torch.rand(n_words, device=device, generator=g)
+    ids = (torch.exp(u * torch.log(torch.tensor(float(vocab_size))))
+           .long() - 1).clamp_(0, vocab_size - 1)
+    sids = torch.arange(n_words, device=device) // sent_len
+    return ids, sids
+
+
+def zipf_counts(vocab_size: int, total_words: int) -> List[int]:
+    """Expected counts for the synthetic corpus (for Sampler/Huffman)."""
+    import numpy as np
+    r = np.arange(1, vocab_size + 1, dtype=np.float64)
+    p = (1.0 / r)
+    p /= p.sum()
+    return list(np.maximum((p * total_words).astype(np.int64), 1))

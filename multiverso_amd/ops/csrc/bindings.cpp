@@ -19,6 +19,9 @@ void mv_launch_row_gather(float*, const float*, const int64_t*, int64_t,
                           int64_t, hipStream_t);
 void mv_launch_row_scatter_add(float*, const float*, const int64_t*, float,
                                int64_t, int64_t, hipStream_t);
+void mv_launch_w2v(float*, float*, float*, float*, const int64_t*, const int*,
+                   const int64_t*, const float*, const int*, float, int64_t,
+                   int64_t, int, hipStream_t);
 }
 
 namespace {
@@ -102,6 +105,37 @@ void row_scatter_add(torch::Tensor shard, torch::Tensor rows,
                             rows.numel(), shard.size(1), cur_stream());
 }
 
+void w2v_train(torch::Tensor in_emb, torch::Tensor out_emb,
+               torch::Tensor in_gsq, torch::Tensor out_gsq,
+               torch::Tensor in_idx, torch::Tensor in_off,
+               torch::Tensor out_idx, torch::Tensor out_label,
+               torch::Tensor out_off, double lr, bool use_adagrad) {
+  check_f32(in_emb, "in_emb"); check_f32(out_emb, "out_emb");
+  check_f32(out_label, "out_label");
+  TORCH_CHECK(in_emb.dim() == 2 && out_emb.dim() == 2, "emb must be 2-D");
+  TORCH_CHECK(in_emb.size(1) == out_emb.size(1), "dim mismatch");
+  TORCH_CHECK(in_emb.size(1) <= 512, "w2v kernel supports dim <= 512");
+  TORCH_CHECK(in_idx.scalar_type() == torch::kInt64 &&
+              out_idx.scalar_type() == torch::kInt64, "idx must be int64");
+  TORCH_CHECK(in_off.scalar_type() == torch::kInt32 &&
+              out_off.scalar_type() == torch::kInt32, "offsets must be int32");
+  int64_t G = in_off.numel() - 1;
+  TORCH_CHECK(out_off.numel() - 1 == G, "group count mismatch");
+  float *igq = nullptr, *ogq = nullptr;
+  if (use_adagrad) {
+    check_f32(in_gsq, "in_gsq"); check_f32(out_gsq, "out_gsq");
+    TORCH_CHECK(in_gsq.sizes() == in_emb.sizes() &&
+                out_gsq.sizes() == out_emb.sizes(), "gsq shape mismatch");
+    igq = in_gsq.data_ptr<float>();
+    ogq = out_gsq.data_ptr<float>();
+  }
+  mv_launch_w2v(in_emb.data_ptr<float>(), out_emb.data_ptr<float>(), igq, ogq,
+                in_idx.data_ptr<int64_t>(), in_off.data_ptr<int>(),
+                out_idx.data_ptr<int64_t>(), out_label.data_ptr<float>(),
+                out_off.data_ptr<int>(), (float)lr, G, in_emb.size(1),
+                use_adagrad ? 1 : 0, cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -113,4 +147,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_gather_out", &row_gather_out, "K6 (preallocated out)");
   m.def("row_scatter_add", &row_scatter_add,
         "K5: shard[rows[i]] += sign*vals[i] (atomic)");
+  m.def("w2v_train", &w2v_train,
+        "K9-K11: fused word2vec block training (skip-gram/CBOW, NS/HS, "
+        "optional adagrad)");
 }

@@ -235,3 +235,132 @@ void mv_launch_row_scatter_add(float* shard, const float* vals,
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// Fused word2vec training kernel (K9+K10+K11):
+// the reference's FeedForward / BPOutputLayer / context-embedding update
+// (Applications/WordEmbedding/src/wordembedding.cpp:57-166) fused into ONE
+// kernel over a block of training groups.
+//
+// A group = one (input set, output set) sample: skip-gram has 1 input and
+// 1+neg outputs (labels 1,0,..); CBOW has window inputs; hierarchical
+// softmax passes labels = 1-code over the Huffman path nodes. The caller
+// maps global word ids to block-local rows of the gathered in_emb/out_emb
+// buffers (the PS "requested parameters" of communicator.cpp:117-155).
+//
+// Geometry: one 64-lane wave per group, grid-stride over groups; each lane
+// owns columns lane, lane+64, ... (DPL = ceil(dim/64) registers). The dot
+// product reduces across the wave with __shfl_xor. Row updates use
+// device-scope atomicAdd: waves share hot vocabulary rows (the reference
+// used unsynchronized hogwild OpenMP updates; atomics keep the sum exact).
+// ADAGRAD variant implements the app-side per-element accumulator math of
+// wordembedding.cpp:101-165 (G += g^2; w += g*lr0/rsqrt(G) when G>1e-10).
+// ---------------------------------------------------------------------------
+
+template <int DPL, bool ADAGRAD>
+__global__ void k_w2v(float* __restrict__ in_emb, float* __restrict__ out_emb,
+                      float* __restrict__ in_gsq, float* __restrict__ out_gsq,
+                      const int64_t* __restrict__ in_idx,
+                      const int* __restrict__ in_off,
+                      const int64_t* __restrict__ out_idx,
+                      const float* __restrict__ out_label,
+                      const int* __restrict__ out_off,
+                      float lr, int G, int dim) {
+  int wid = (int)((blockIdx.x * (int64_t)blockDim.x + threadIdx.x) >> 6);
+  int lane = threadIdx.x & 63;
+  int nwaves = (int)((gridDim.x * (int64_t)blockDim.x) >> 6);
+  for (int g = wid; g < G; g += nwaves) {
+    float h[DPL], err[DPL];
+#pragma unroll
+    for (int d = 0; d < DPL; ++d) { h[d] = 0.f; err[d] = 0.f; }
+    int ib = in_off[g], ie = in_off[g + 1];
+    for (int i = ib; i < ie; ++i) {
+      const float* row = in_emb + in_idx[i] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) h[d] += row[c];
+      }
+    }
+    if (ie - ib > 1) {
+      float inv = 1.f / (float)(ie - ib);
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) h[d] *= inv;
+    }
+    int ob = out_off[g], oe = out_off[g + 1];
+    for (int o = ob; o < oe; ++o) {
+      float* w = out_emb + out_idx[o] * dim;
+      float wv[DPL];
+      float f = 0.f;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        wv[d] = (c < dim) ? w[c] : 0.f;
+        f += h[d] * wv[d];
+      }
+#pragma unroll
+      for (int s = 32; s; s >>= 1) f += __shfl_xor(f, s, 64);
+      f = 1.f / (1.f + expf(-f));
+      float e = out_label[o] - f;  // NS: label-f; HS: caller passes 1-code
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) {
+          err[d] += e * wv[d];  // hidden_err uses the PRE-update classifier
+          if (ADAGRAD) {
+            float gg = e * h[d];
+            float* gq = out_gsq + out_idx[o] * dim + c;
+            float G2 = atomicAdd(gq, gg * gg) + gg * gg;
+            if (G2 > 1e-10f) atomicAdd(&w[c], gg * lr * __frsqrt_rn(G2));
+          } else {
+            atomicAdd(&w[c], e * lr * h[d]);
+          }
+        }
+      }
+    }
+    for (int i = ib; i < ie; ++i) {
+      float* row = in_emb + in_idx[i] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) {
+          if (ADAGRAD) {
+            float* gq = in_gsq + in_idx[i] * dim + c;
+            float G2 = atomicAdd(gq, err[d] * err[d]) + err[d] * err[d];
+            if (G2 > 1e-10f) atomicAdd(&row[c], err[d] * lr * __frsqrt_rn(G2));
+          } else {
+            atomicAdd(&row[c], lr * err[d]);
+          }
+        }
+      }
+    }
+  }
+}
+
+extern "C" void mv_launch_w2v(float* in_emb, float* out_emb,
+                              float* in_gsq, float* out_gsq,
+                              const int64_t* in_idx, const int* in_off,
+                              const int64_t* out_idx, const float* out_label,
+                              const int* out_off, float lr, int64_t G,
+                              int64_t dim, int use_adagrad, hipStream_t s) {
+  if (!G) return;
+  int grid = grid_for(G * 64);  // 4 waves per 256-thread block
+  int dpl = (int)((dim + 63) / 64);
+#define W2V_CASE(D)                                                          \
+  case D:                                                                    \
+    if (use_adagrad)                                                         \
+      k_w2v<D, true><<<grid, BLOCK, 0, s>>>(in_emb, out_emb, in_gsq,         \
+          out_gsq, in_idx, in_off, out_idx, out_label, out_off, lr, (int)G,  \
+          (int)dim);                                                         \
+    else                                                                     \
+      k_w2v<D, false><<<grid, BLOCK, 0, s>>>(in_emb, out_emb, in_gsq,        \
+          out_gsq, in_idx, in_off, out_idx, out_label, out_off, lr, (int)G,  \
+          (int)dim);                                                         \
+    break;
+  switch (dpl) {
+    W2V_CASE(1) W2V_CASE(2) W2V_CASE(3) W2V_CASE(4)
+    W2V_CASE(5) W2V_CASE(6) W2V_CASE(7) W2V_CASE(8)
+    default: break;  // dim > 512 unsupported by this kernel
+  }
+#undef W2V_CASE
+}
