@@ -51,6 +51,11 @@ class WordEmbeddingOption:
     data_block_size: int = 100_000  # words per block (ref default 1e6)
     unigram_table_size: int = 10_000_000
     seed: int = 1
+    # Within a kernel launch, groups train hogwild-concurrently (row reads
+    # are pre-update, like the reference's unsynchronized OpenMP trainers).
+    # Duplicate-heavy blocks are chunked into sequential launches so later
+    # occurrences of a pair see earlier updates — bounded staleness.
+    max_groups_per_launch: int = 1 << 20
 
 
 class WordEmbedding:
@@ -279,12 +284,23 @@ class WordEmbedding:
             from ... import ops
             hip = ops.module(required=True)
             dummy = in_buf  # unused when not adagrad
-            hip.w2v_train(in_buf, out_buf,
-                          in_gsq if in_gsq is not None else dummy,
-                          out_gsq if out_gsq is not None else dummy,
-                          in_local, in_off.to(torch.int32),
-                          out_local, out_label, out_off.to(torch.int32),
-                          self.learning_rate, self.opt.use_adagrad)
+            igq = in_gsq if in_gsq is not None else dummy
+            ogq = out_gsq if out_gsq is not None else dummy
+            in_off = in_off.to(torch.int32)
+            out_off = out_off.to(torch.int32)
+            g_total = in_off.numel() - 1
+            step = self.opt.max_groups_per_launch
+            for g0 in range(0, g_total, step):
+                g1 = min(g0 + step, g_total)
+                i0, i1 = int(in_off[g0]), int(in_off[g1])
+                o0, o1 = int(out_off[g0]), int(out_off[g1])
+                hip.w2v_train(in_buf, out_buf, igq, ogq,
+                              in_local[i0:i1].contiguous(),
+                              (in_off[g0:g1 + 1] - i0).contiguous(),
+                              out_local[o0:o1].contiguous(),
+                              out_label[o0:o1].contiguous(),
+                              (out_off[g0:g1 + 1] - o0).contiguous(),
+                              self.learning_rate, self.opt.use_adagrad)
         else:
             _w2v_train_torch(in_buf, out_buf, in_gsq, out_gsq, in_local,
                              in_off, out_local, out_label, out_off,

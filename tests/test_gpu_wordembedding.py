@@ -77,9 +77,12 @@ def test_w2v_gpu_end_to_end_learns():
                                                          WordEmbeddingOption)
     mv.init()
     torch.manual_seed(0)
+    # duplicate-heavy corpus: chunk launches small so later duplicates see
+    # earlier updates (sequential-equivalent semantics)
     opt = WordEmbeddingOption(embedding_size=64, window=1, negative_num=5,
                               init_learning_rate=0.1,
-                              total_words=10_000_000, seed=3)
+                              total_words=10_000_000, seed=3,
+                              max_groups_per_launch=64)
     model = WordEmbedding(opt, [100] * 20)
     words = torch.stack([torch.arange(0, 20, 2).repeat(100),
                          torch.arange(1, 20, 2).repeat(100)],
