@@ -26,7 +26,10 @@ import torch
 from .configure import define_flag, get_flag, parse_cmd_flags, set_flag
 from .dashboard import Dashboard, Monitor, Timer, monitor
 from .log import CHECK, CHECK_NOTNULL, FatalError, LogLevel, log
-from .tables import ArrayTable, KVTable, MatrixTable
+from .async_buffer import ASyncBuffer
+from .checkpoint import checkpoint, restore
+from .io import Stream, StreamFactory, TextReader, URI
+from .tables import ArrayTable, KVTable, MatrixTable, SparseMatrixTable
 from .updaters import AddOption
 from .zoo import Role, Zoo
 
@@ -165,8 +168,10 @@ class MatrixTableHandler:
 __all__ = [
     "init", "shutdown", "barrier", "workers_num", "servers_num", "worker_id",
     "server_id", "is_master_worker", "rank", "size", "aggregate",
-    "ArrayTable", "MatrixTable", "KVTable",
+    "ArrayTable", "MatrixTable", "SparseMatrixTable", "KVTable",
     "ArrayTableHandler", "MatrixTableHandler",
+    "checkpoint", "restore", "ASyncBuffer",
+    "Stream", "StreamFactory", "TextReader", "URI",
     "AddOption", "set_flag", "get_flag", "define_flag", "parse_cmd_flags",
     "Dashboard", "Monitor", "Timer", "monitor",
     "log", "LogLevel", "CHECK", "CHECK_NOTNULL", "FatalError",

@@ -1,0 +1,119 @@
+"""SparseMatrixTable — MatrixTable with stale-aware (freshness-filtered)
+whole-table Get.
+
+Capability parity with the reference SparseMatrixTable
+(src/table/sparse_matrix_table.cpp) / unified Matrix with is_sparse
+(src/table/matrix.cpp): the server keeps a per-(worker, row) up-to-date
+bitmap; a worker's Get returns only the rows that are stale FOR THAT
+WORKER (UpdateGetState, :226-258), and an Add invalidates the touched rows
+for every OTHER worker (UpdateAddState, :200-223). Workers therefore keep
+a local cache of the whole table and only stale rows travel.
+
+MI355X mapping: the bitmap lives with the owning shard as a bool tensor in
+HBM; ``get_into(cache)`` is a collective where each owner builds each
+requester's stale row list (a masked select), serves those rows (K6
+gather) and marks them fresh; the rows travel in one all-to-all. Keyed
+adds travel through the base class and invalidate per source rank."""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from ..comm import all_to_all_values
+from ..dashboard import monitor
+from ..log import CHECK
+from ..updaters import AddOption
+from .matrix_table import MatrixTable
+
+
+class SparseMatrixTable(MatrixTable):
+    def __init__(self, num_row: int, num_col: int,
+                 dtype: torch.dtype = torch.float32,
+                 updater_type: Optional[str] = None,
+                 random_init=None) -> None:
+        super().__init__(num_row, num_col, dtype, updater_type, random_init)
+        nw = self.zoo.num_workers
+        # up_to_date[w, r] == True -> worker w has the current row r
+        self.up_to_date = torch.zeros(nw, self.local_rows, dtype=torch.bool,
+                                      device=self.device)
+
+    # ---- invalidation hooks ----
+    def _invalidate_all_but(self, local_ids: torch.Tensor,
+                            worker: Optional[int]) -> None:
+        self.up_to_date[:, local_ids] = False
+        if worker is not None:
+            self.up_to_date[worker, local_ids] = True
+
+    def add(self, delta, option=None, async_op: bool = False):
+        h = super().add(delta, option, async_op=async_op)
+        # whole-table add touches every row; a dense add comes from every
+        # worker at once (reduce-scatter), so everyone is stale.
+        self.up_to_date.zero_()
+        return h
+
+    def add_rows(self, row_ids, values, option: Optional[AddOption] = None,
+                 source_worker: Optional[int] = None) -> None:
+        ids = torch.as_tensor(row_ids, dtype=torch.int64, device=self.device)
+        vals = values.to(self.device, self.dtype).contiguous()
+        from ..comm import all_to_all_rows
+        with monitor("worker.add_rows"):
+            in_ids, in_vals, recv_sizes, _ = all_to_all_rows(
+                ids, vals.view(-1), self.spec, self.num_col)
+            if in_ids.numel():
+                local = self._local_rows_of(in_ids)
+                with monitor("server.update_rows"):
+                    self._scatter_update_local(
+                        local, in_vals.view(-1, self.num_col), option)
+                # UpdateAddState: conservatively invalidate the touched
+                # rows for EVERY worker (the reference spares the adding
+                # worker, whose cache is pre-add anyway; invalidating all
+                # is never-stale-safe and costs one extra row re-pull)
+                self.up_to_date[:, local] = False
+
+    # ---- stale-filtered whole-table get ----
+    def get_into(self, cache: torch.Tensor) -> int:
+        """Overwrite only the rows of ``cache`` that are stale for this
+        worker; marks them fresh. Returns the number of rows received.
+        Collective: every rank must call."""
+        CHECK(cache.shape == (self.num_row, self.num_col),
+              "cache must be the full table shape")
+        self.flush()
+        me = self.zoo.worker_id
+        nw = self.zoo.num_workers
+        with monitor("worker.sparse_get"):
+            if self.zoo.size == 1:
+                stale = (~self.up_to_date[0]).nonzero().reshape(-1)
+                if stale.numel():
+                    cache[stale] = self._gather_local(stale)
+                    self.up_to_date[0, stale] = True
+                return int(stale.numel())
+            # 1) each owner builds per-requester stale lists
+            stale_lists = [(~self.up_to_date[w]).nonzero().reshape(-1)
+                           for w in range(nw)]
+            send_rows = [int(s.numel()) for s in stale_lists]
+            # 2) exchange counts, then ids+values
+            cnt = torch.tensor(send_rows, dtype=torch.int64,
+                               device=self.device)
+            rcnt = torch.empty(nw, dtype=torch.int64, device=self.device)
+            dist.all_to_all_single(rcnt, cnt)
+            recv_rows = rcnt.tolist()
+            all_ids = torch.cat(stale_lists) + self.row_offset
+            got_ids = torch.empty(sum(recv_rows), dtype=torch.int64,
+                                  device=self.device)
+            dist.all_to_all_single(got_ids, all_ids, recv_rows, send_rows)
+            served = (self._gather_local(torch.cat(stale_lists))
+                      if all_ids.numel() else
+                      torch.empty(0, self.num_col, dtype=self.dtype,
+                                  device=self.device))
+            got_vals = all_to_all_values(served.view(-1), send_rows,
+                                         recv_rows, self.num_col)
+            # 3) write into cache; mark fresh on the server side
+            if got_ids.numel():
+                cache[got_ids] = got_vals.view(-1, self.num_col)
+            for w in range(nw):
+                if send_rows[w]:
+                    self.up_to_date[w, stale_lists[w]] = True
+            return int(got_ids.numel())

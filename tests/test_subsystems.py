@@ -1,0 +1,152 @@
+"""Tests for io streams, checkpoint driver, ASyncBuffer, SparseFilter,
+and the stale-aware SparseMatrixTable (single- and multi-process)."""
+
+import time
+
+import pytest
+import torch
+
+import multiverso_amd as mv
+from conftest import run_dist
+
+
+@pytest.fixture()
+def env():
+    mv.init()
+    yield
+    mv.shutdown()
+
+
+# ---- io ----
+
+def test_local_stream_roundtrip(tmp_path):
+    p = f"file://{tmp_path}/x.bin"
+    with mv.StreamFactory.get_stream(p, "w") as s:
+        s.write(b"hello\nworld\n")
+    with mv.StreamFactory.get_stream(p, "r") as s:
+        assert s.read() == b"hello\nworld\n"
+    r = mv.TextReader(p)
+    assert r.get_line() == "hello"
+    assert r.get_line() == "world"
+    assert r.get_line() is None
+
+
+def test_uri_parse():
+    u = mv.URI("hdfs://nn/path/f")
+    assert u.scheme == "hdfs" and u.path == "nn/path/f"
+    assert mv.URI("/plain/path").scheme == "file"
+    with pytest.raises(NotImplementedError):
+        mv.StreamFactory.get_stream("hdfs://x/y", "r")
+    with pytest.raises(ValueError):
+        mv.StreamFactory.get_stream("s3://x/y", "r")
+
+
+# ---- checkpoint driver ----
+
+def test_checkpoint_restore_driver(env, tmp_path):
+    a = mv.ArrayTable(16)
+    m = mv.MatrixTable(4, 4)
+    a.add(torch.arange(16, dtype=torch.float32))
+    m.add(torch.ones(4, 4))
+    mv.checkpoint(str(tmp_path / "ckpt"))
+    mv.shutdown()
+    mv.init()
+    a2 = mv.ArrayTable(16)
+    m2 = mv.MatrixTable(4, 4)
+    mv.restore(str(tmp_path / "ckpt"))
+    assert torch.equal(a2.get(), torch.arange(16, dtype=torch.float32))
+    assert torch.equal(m2.get(), torch.ones(4, 4))
+
+
+# ---- ASyncBuffer ----
+
+def test_async_buffer_prefetch():
+    calls = []
+
+    def fill(buf):
+        time.sleep(0.01)
+        buf.append(len(calls))
+        calls.append(1)
+
+    ab = mv.ASyncBuffer([], [], fill)
+    b0 = ab.get()
+    assert b0 == [0]
+    b1 = ab.get()
+    assert b1 == [1]
+    b0b = ab.get()
+    assert b0b[0] == 0 and len(b0b) == 2  # same buffer object, refilled
+
+
+# ---- sparse filter ----
+
+def test_sparse_filter_roundtrip():
+    from multiverso_amd import sparse_filter as sf
+    v = torch.zeros(1000)
+    v[[3, 500, 999]] = torch.tensor([1.0, -2.0, 3.5])
+    payload, comp = sf.filter_in(v)
+    assert comp and payload.numel() == 1 + 6
+    assert torch.equal(sf.filter_out(payload, comp, 1000), v)
+    dense = torch.ones(10)
+    payload, comp = sf.filter_in(dense)
+    assert not comp and torch.equal(payload, dense)
+
+
+# ---- sparse matrix table ----
+
+def test_sparse_matrix_local(env):
+    t = mv.SparseMatrixTable(8, 2)
+    cache = torch.full((8, 2), -1.0)
+    assert t.get_into(cache) == 8      # everything stale initially
+    assert torch.equal(cache, torch.zeros(8, 2))
+    assert t.get_into(cache) == 0      # now fresh
+    t.add_rows([2, 6], torch.ones(2, 2))
+    got = t.get_into(cache)
+    assert got == 2
+    assert torch.equal(cache[2], torch.ones(2))
+    assert torch.equal(cache[0], torch.zeros(2))
+
+
+def _sparse_matrix_dist(rank, world):
+    import multiverso_amd as mv
+    import torch
+    mv.init(sync=True)
+    t = mv.SparseMatrixTable(9, 3)
+    cache = torch.zeros(9, 3)
+    n = t.get_into(cache)
+    assert n == 9, n
+    assert t.get_into(cache) == 0
+    # rank 0 adds row 1; both ranks then see it stale and re-pull
+    if rank == 0:
+        t.add_rows([1], torch.ones(1, 3))
+    else:
+        t.add_rows([], torch.zeros(0, 3))
+    n = t.get_into(cache)
+    assert n == 1, (rank, n)
+    assert torch.equal(cache[1], torch.ones(3))
+    mv.shutdown()
+
+
+def test_sparse_matrix_dist():
+    run_dist(_sparse_matrix_dist, 2)
+
+
+def _checkpoint_dist(rank, world):
+    import multiverso_amd as mv
+    import torch
+    import tempfile, os
+    mv.init(sync=True)
+    d = os.path.join(tempfile.gettempdir(), "mv_ckpt_test")
+    a = mv.ArrayTable(10)
+    a.add(torch.ones(10))
+    mv.checkpoint(d)
+    mv.shutdown()
+    mv.init(sync=True)
+    a2 = mv.ArrayTable(10)
+    mv.restore(d)
+    got = a2.get()
+    assert torch.equal(got, torch.full((10,), float(world))), got
+    mv.shutdown()
+
+
+def test_checkpoint_dist():
+    run_dist(_checkpoint_dist, 2)
