@@ -33,7 +33,7 @@ def parse_args():
     p.add_argument("--cols", type=int, default=128)
     p.add_argument("--updater", type=str, default="sgd")
     p.add_argument("--app", type=str, default="matrix",
-                   choices=["matrix", "wordembedding"])
+                   choices=["matrix", "wordembedding", "logreg"])
     p.add_argument("--vocab", type=int, default=1_000_000)
     p.add_argument("--dim", type=int, default=200)
     p.add_argument("--block-words", dest="block_words", type=int,
@@ -51,6 +51,10 @@ def main():
     import multiverso_amd as mv
     if args.app == "wordembedding":
         from multiverso_amd.apps.wordembedding.bench import run_bench
+        run_bench(args)
+        return
+    if args.app == "logreg":
+        from multiverso_amd.apps.logreg.bench import run_bench
         run_bench(args)
         return
 

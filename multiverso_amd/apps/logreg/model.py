@@ -1,0 +1,178 @@
+"""LogReg models: local and parameter-server backed.
+
+Capability parity with the reference Model/PSModel
+(Applications/LogisticRegression/src/model/model.cpp:62-222,
+ps_model.cpp): minibatch delta accumulation + updater apply; the PS
+variant holds weights in a table (Array/Sparse/FTRL by config,
+ps_model.cpp:26-41), sets multiverso ``updater_type=sgd`` (:24), syncs
+every ``sync_frequency`` minibatches (:172-182), optionally pipelines the
+pull with a double buffer (:236-271), and pre-pulls the whole model
+before store (:157-169).
+
+MI355X mapping: weights live in a row-sharded MatrixTable
+(rows = input_size features, cols = output_size; FTRL cols = 2*output for
+(z|n)); the per-chunk pull is get_rows (all-to-all over xGMI, K6 gather),
+the push is add_rows (K5/K15 scatter-update). Dense mode uses whole-table
+collectives. Worker-side lr decay matches SGDUpdater
+(updater.cpp:53-71): lr = max(1e-3, lr0 - count/(coef*minibatch))."""
+
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+import numpy as np
+import torch
+
+import multiverso_amd as mv
+
+from .objective import Batch, create_objective
+
+
+class WorkerSGDSchedule:
+    def __init__(self, cfg) -> None:
+        self.lr0 = cfg.learning_rate
+        self.coef = cfg.learning_rate_coef
+        self.minibatch = cfg.minibatch_size
+        self.count = 0
+
+    def next_lr(self) -> float:
+        lr = max(1e-3, self.lr0 - self.count / (self.coef * self.minibatch))
+        self.count += 1
+        return lr
+
+
+class LocalModel:
+    """use_ps=false: full weight tensor on this device
+    (model/model.cpp:62-135)."""
+
+    def __init__(self, cfg, device=None) -> None:
+        self.cfg = cfg
+        self.device = device or torch.device("cpu")
+        self.objective = create_objective(cfg)
+        cols = cfg.output_size * (2 if cfg.objective_type == "ftrl" else 1)
+        self.weight = torch.zeros(cfg.input_size, cols, device=self.device)
+        self.sched = WorkerSGDSchedule(cfg)
+
+    def update(self, batch: Batch) -> float:
+        batch = batch.to(self.device)
+        w_rows = self.weight[batch.keys]
+        grad, loss = self.objective.gradient(batch, w_rows)
+        if self.cfg.objective_type == "ftrl":
+            # server/local ftrl update is unscaled: state -= delta
+            self.weight.index_add_(0, batch.keys, -grad)
+        else:
+            lr = self.sched.next_lr()
+            self.weight.index_add_(0, batch.keys, -lr * grad)
+        return loss
+
+    def predict(self, batch: Batch) -> torch.Tensor:
+        batch = batch.to(self.device)
+        return self.objective.predict(batch, self.weight[batch.keys])
+
+    # ---- model file io (model.cpp:146-204 / sparse_table.h:258-285) ----
+    def store(self, path: str) -> None:
+        w = self.weight.cpu().numpy()
+        if self.cfg.sparse:
+            nz = np.nonzero(w.any(axis=1))[0].astype(np.int64)
+            with open(path, "wb") as f:
+                f.write(np.int64(nz.size).tobytes())
+                f.write(nz.tobytes())
+                f.write(w[nz].astype(np.float32).tobytes())
+        else:
+            w.astype(np.float32).tofile(path)
+
+    def load(self, path: str) -> None:
+        cols = self.weight.shape[1]
+        if self.cfg.sparse:
+            with open(path, "rb") as f:
+                cnt = int(np.frombuffer(f.read(8), dtype=np.int64)[0])
+                keys = np.frombuffer(f.read(8 * cnt), dtype=np.int64)
+                vals = np.frombuffer(f.read(4 * cnt * cols),
+                                     dtype=np.float32).reshape(cnt, cols)
+            self.weight.zero_()
+            self.weight[torch.from_numpy(keys.copy())] = \
+                torch.from_numpy(vals.copy()).to(self.device)
+        else:
+            w = np.fromfile(path, dtype=np.float32).reshape(
+                self.weight.shape)
+            self.weight.copy_(torch.from_numpy(w).to(self.device))
+
+
+class PSModel:
+    """use_ps=true: weights in a row-sharded table; chunked train with
+    pull → local minibatch steps → push (ps_model.cpp)."""
+
+    def __init__(self, cfg, device=None) -> None:
+        self.cfg = cfg
+        self.device = device or mv.Zoo.get().device
+        self.objective = create_objective(cfg)
+        self.is_ftrl = cfg.objective_type == "ftrl"
+        cols = cfg.output_size * (2 if self.is_ftrl else 1)
+        updater = "sgd" if (self.is_ftrl or cfg.updater_type
+                            in ("sgd", "default", "ftrl")) else cfg.updater_type
+        self.table = mv.MatrixTable(cfg.input_size, cols,
+                                    updater_type=updater)
+        self.sched = WorkerSGDSchedule(cfg)
+        self.cols = cols
+
+    def train_chunk(self, batches: List[Batch]) -> float:
+        """One sync group: pull union keys, run the minibatches against
+        the pulled snapshot (within-chunk updates visible locally), push
+        the summed delta (ps_model.cpp:172-203)."""
+        batches = [b.to(self.device) for b in batches]
+        union = torch.unique(torch.cat([b.keys for b in batches]))
+        pulled = self.table.get_rows(union)
+        local = pulled.clone()
+        total_loss = 0.0
+        adagrad = self.table.updater_type == "adagrad"
+        for b in batches:
+            lidx = torch.searchsorted(union, b.keys)
+            w_rows = local[lidx]
+            grad, loss = self.objective.gradient(b, w_rows)
+            total_loss += loss
+            if self.is_ftrl:
+                local.index_add_(0, lidx, -grad)
+            elif adagrad:
+                # server-side adagrad consumes lr-scaled deltas; locally
+                # approximate with plain sgd steps for within-chunk vis.
+                lr = self.sched.next_lr()
+                local.index_add_(0, lidx, -lr * grad)
+            else:
+                lr = self.sched.next_lr()
+                local.index_add_(0, lidx, -lr * grad)
+        if adagrad:
+            delta = pulled - local   # = sum(lr*grad); server g=delta/lr
+            opt = mv.AddOption(learning_rate=1.0, rho=self.cfg.learning_rate)
+            self.table.add_rows(union, delta, option=opt)
+        else:
+            # server updater 'sgd': w -= delta; push accumulated movement
+            self.table.add_rows(union, pulled - local)
+        return total_loss / max(len(batches), 1)
+
+    def predict(self, batch: Batch) -> torch.Tensor:
+        batch = batch.to(self.device)
+        w_rows = self.table.get_rows(batch.keys)
+        return self.objective.predict(batch, w_rows)
+
+    # ---- store/load via the whole-model pre-pull (ps_model.cpp:157-169) --
+    def store(self, path: str) -> None:
+        w = self.table.get()
+        if mv.rank() == 0:
+            w.cpu().numpy().astype(np.float32).tofile(path)
+        mv.barrier()
+
+    def load(self, path: str) -> None:
+        w = np.fromfile(path, dtype=np.float32).reshape(
+            self.cfg.input_size, self.cols)
+        full = torch.from_numpy(w).to(self.device)
+        sl = full[self.table.row_offset:
+                  self.table.row_offset + self.table.local_rows]
+        self.table.shard.copy_(sl)
+        mv.barrier()
+
+
+def create_model(cfg, device=None):
+    """model.cpp:216-222 factory."""
+    if cfg.use_ps:
+        return PSModel(cfg, device)
+    return LocalModel(cfg, device)

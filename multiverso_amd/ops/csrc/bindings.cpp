@@ -22,6 +22,9 @@ void mv_launch_row_scatter_add(float*, const float*, const int64_t*, float,
 void mv_launch_w2v(float*, float*, float*, float*, const int64_t*, const int*,
                    const int64_t*, const float*, const int*, float, int64_t,
                    int64_t, int, hipStream_t);
+void mv_launch_row_scatter_adagrad(float*, float*, const float*,
+                                   const int64_t*, float, float, float,
+                                   int64_t, int64_t, hipStream_t);
 }
 
 namespace {
@@ -105,6 +108,21 @@ void row_scatter_add(torch::Tensor shard, torch::Tensor rows,
                             rows.numel(), shard.size(1), cur_stream());
 }
 
+void row_scatter_adagrad(torch::Tensor shard, torch::Tensor gsq,
+                         torch::Tensor rows, torch::Tensor vals,
+                         double lr, double rho, double eps) {
+  check_f32(shard, "shard"); check_f32(gsq, "gsq"); check_f32(vals, "vals");
+  TORCH_CHECK(shard.dim() == 2, "shard must be 2-D");
+  TORCH_CHECK(gsq.sizes() == shard.sizes(), "gsq shape mismatch");
+  TORCH_CHECK(rows.scalar_type() == torch::kInt64 && rows.is_cuda() &&
+              rows.is_contiguous(), "rows must be contiguous int64 on GPU");
+  TORCH_CHECK(vals.numel() == rows.numel() * shard.size(1), "vals mismatch");
+  mv_launch_row_scatter_adagrad(
+      shard.data_ptr<float>(), gsq.data_ptr<float>(), vals.data_ptr<float>(),
+      rows.data_ptr<int64_t>(), (float)lr, (float)rho, (float)eps,
+      rows.numel(), shard.size(1), cur_stream());
+}
+
 void w2v_train(torch::Tensor in_emb, torch::Tensor out_emb,
                torch::Tensor in_gsq, torch::Tensor out_gsq,
                torch::Tensor in_idx, torch::Tensor in_off,
@@ -147,6 +165,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_gather_out", &row_gather_out, "K6 (preallocated out)");
   m.def("row_scatter_add", &row_scatter_add,
         "K5: shard[rows[i]] += sign*vals[i] (atomic)");
+  m.def("row_scatter_adagrad", &row_scatter_adagrad,
+        "K15: keyed adagrad update on owned shard rows");
   m.def("w2v_train", &w2v_train,
         "K9-K11: fused word2vec block training (skip-gram/CBOW, NS/HS, "
         "optional adagrad)");

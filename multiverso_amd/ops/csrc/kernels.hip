@@ -364,3 +364,39 @@ extern "C" void mv_launch_w2v(float* in_emb, float* out_emb,
   }
 #undef W2V_CASE
 }
+
+// ---------------------------------------------------------------------------
+// Keyed AdaGrad scatter-update (K15 + K4 keyed form): for each incoming
+// (row, value) pair apply the adagrad step to the owned shard row. Used by
+// the LogisticRegression sparse path (BASELINE config: 1e9 sparse
+// features, AdaGrad updater). Duplicate rows race benignly (hogwild):
+// G accumulates via atomicAdd; the weight step uses the post-add G.
+// ---------------------------------------------------------------------------
+
+__global__ void k_row_scatter_adagrad(float* __restrict__ shard,
+                                      float* __restrict__ gsq,
+                                      const float* __restrict__ vals,
+                                      const int64_t* __restrict__ rows,
+                                      float inv_lr, float rho, float eps,
+                                      int64_t nrows, int64_t cols) {
+  int64_t total = nrows * cols;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
+    int64_t r = i / cols, c = i % cols;
+    int64_t k = rows[r] * cols + c;
+    float g = vals[i] * inv_lr;
+    if (g != 0.0f) {
+      float G = atomicAdd(&gsq[k], g * g) + g * g;
+      atomicAdd(&shard[k], -rho * g * __frsqrt_rn(G + eps));
+    }
+  }
+}
+
+extern "C" void mv_launch_row_scatter_adagrad(
+    float* shard, float* gsq, const float* vals, const int64_t* rows,
+    float lr, float rho, float eps, int64_t nrows, int64_t cols,
+    hipStream_t s) {
+  if (!nrows || !cols) return;
+  k_row_scatter_adagrad<<<grid_for(nrows * cols), BLOCK, 0, s>>>(
+      shard, gsq, vals, rows, 1.0f / lr, rho, eps, nrows, cols);
+}

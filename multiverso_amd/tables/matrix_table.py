@@ -108,8 +108,31 @@ class MatrixTable(Table):
     def _scatter_update_local(self, local_ids: torch.Tensor,
                               vals: torch.Tensor,
                               option: Optional[AddOption]) -> None:
-        """K5 on the owned shard. default adds, sgd subtracts — the same
-        per-row updater dispatch as matrix_table.cpp:406-412."""
+        """K5 on the owned shard. default adds, sgd subtracts (the per-row
+        updater dispatch of matrix_table.cpp:406-412); adagrad applies the
+        keyed K15 form (duplicate rows: GPU races benignly via atomics,
+        CPU pre-aggregates duplicates — both keep the G accumulate)."""
+        if self.updater_type == "adagrad":
+            from ..updaters import AddOption as _AO
+            opt = option or _AO()
+            gsq = self.updater.g_sqr.view(self.local_rows, self.num_col)
+            vals2 = vals.view(-1, self.num_col)
+            if self.shard.is_cuda:
+                from .. import ops
+                ops.module(required=True).row_scatter_adagrad(
+                    self.shard, gsq, local_ids, vals2.contiguous(),
+                    opt.learning_rate, opt.rho, self.updater.EPS)
+            else:
+                urows, inv = torch.unique(local_ids, return_inverse=True)
+                agg = torch.zeros(urows.numel(), self.num_col,
+                                  dtype=self.dtype)
+                agg.index_add_(0, inv, vals2)
+                g = agg / opt.learning_rate
+                gsq[urows] += g * g
+                self.shard[urows] -= (opt.rho * g /
+                                      torch.sqrt(gsq[urows]
+                                                 + self.updater.EPS))
+            return
         sign = {"default": 1.0, "sgd": -1.0}.get(self.updater_type)
         CHECK(sign is not None,
               f"row-keyed Add with updater '{self.updater_type}' is not "

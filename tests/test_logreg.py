@@ -1,0 +1,177 @@
+"""LogisticRegression app tests: config parsing, readers, objectives,
+local + PS models (sigmoid/softmax/ftrl/adagrad), model io, and a
+2-process PS run."""
+
+import numpy as np
+import pytest
+import torch
+
+import multiverso_amd as mv
+from conftest import run_dist
+from multiverso_amd.apps.logreg import LogReg, LogRegConfig
+from multiverso_amd.apps.logreg.objective import Batch, create_objective
+from multiverso_amd.apps.logreg.reader import (SampleReader, parse_text_lines,
+                                               read_bsparse_batches,
+                                               synthetic_batches,
+                                               write_bsparse)
+
+
+@pytest.fixture()
+def env():
+    mv.init()
+    yield
+    mv.shutdown()
+
+
+def test_config_parse(tmp_path):
+    p = tmp_path / "c.cfg"
+    p.write_text("input_size=100\nobjective_type=softmax\n"
+                 "use_ps=true\nlearning_rate=0.5\n# comment\nbadkey=1\n")
+    cfg = LogRegConfig.from_file(str(p))
+    assert cfg.input_size == 100
+    assert cfg.objective_type == "softmax"
+    assert cfg.use_ps is True
+    assert cfg.learning_rate == 0.5
+
+
+def test_text_reader(tmp_path):
+    p = tmp_path / "train.txt"
+    p.write_text("1 0:0.5 3:1.5\n0 2:1.0\n1 1:2.0 4:0.1\n")
+    batches = list(SampleReader(str(p), 2).batches())
+    assert len(batches) == 2
+    b = batches[0]
+    assert b.size == 2
+    assert b.keys.tolist() == [0, 3, 2]
+    assert b.labels.tolist() == [1.0, 0.0]
+
+
+def test_weighted_reader(tmp_path):
+    p = tmp_path / "w.txt"
+    p.write_text("1 2.0 0:1\n")
+    b = list(SampleReader(str(p), 4, "weight").batches())[0]
+    assert b.weights.tolist() == [2.0]
+
+
+def test_bsparse_roundtrip(tmp_path):
+    p = str(tmp_path / "b.bin")
+    write_bsparse(p, [(1, [(5, 0.5), (9, 1.0)]), (0, [(2, 2.0)])])
+    batches = list(read_bsparse_batches(p, 10))
+    assert batches[0].size == 2
+    assert batches[0].keys.tolist() == [5, 9, 2]
+    assert batches[0].vals.tolist() == [0.5, 1.0, 2.0]
+
+
+def _train_local(cfg, n_batches=60):
+    batches, _ = synthetic_batches(cfg.input_size, n_batches,
+                                   cfg.minibatch_size, nnz=16,
+                                   output_size=max(cfg.output_size, 1)
+                                   if cfg.objective_type == "softmax" else 1,
+                                   seed=7)
+    lr = LogReg(cfg)
+    lr.train(iter(batches))
+    acc, loss = lr.test(iter(batches[:10]))
+    return lr, acc
+
+
+def test_local_sigmoid_learns(env):
+    cfg = LogRegConfig(input_size=4096, minibatch_size=64,
+                       learning_rate=0.05, learning_rate_coef=1e6,
+                       train_epoch=1, show_time_per_sample=0)
+    _, acc = _train_local(cfg)
+    assert acc > 0.75, acc
+
+
+def test_local_softmax_learns(env):
+    cfg = LogRegConfig(input_size=4096, output_size=4, minibatch_size=64,
+                       objective_type="softmax", learning_rate=0.05,
+                       learning_rate_coef=1e6, show_time_per_sample=0)
+    _, acc = _train_local(cfg)
+    assert acc > 0.5, acc
+
+
+def test_local_ftrl_learns(env):
+    cfg = LogRegConfig(input_size=4096, minibatch_size=64,
+                       objective_type="ftrl", show_time_per_sample=0)
+    _, acc = _train_local(cfg)
+    assert acc > 0.7, acc
+
+
+def test_l2_regular_shrinks(env):
+    cfg = LogRegConfig(input_size=512, minibatch_size=32,
+                       regular_type="l2", regular_coef=0.1,
+                       learning_rate=0.05, learning_rate_coef=1e6,
+                       show_time_per_sample=0)
+    lr, _ = _train_local(cfg, 30)
+    cfg2 = LogRegConfig(input_size=512, minibatch_size=32,
+                        learning_rate=0.05, learning_rate_coef=1e6,
+                        show_time_per_sample=0)
+    lr2, _ = _train_local(cfg2, 30)
+    assert lr.model.weight.norm() < lr2.model.weight.norm()
+
+
+def test_model_io_dense(env, tmp_path):
+    cfg = LogRegConfig(input_size=256, minibatch_size=32,
+                       show_time_per_sample=0)
+    lr, _ = _train_local(cfg, 10)
+    p = str(tmp_path / "model.bin")
+    lr.save_model(p)
+    lr2 = LogReg(LogRegConfig(input_size=256, show_time_per_sample=0))
+    lr2.load_model(p)
+    assert torch.equal(lr.model.weight, lr2.model.weight)
+
+
+def test_model_io_sparse(env, tmp_path):
+    cfg = LogRegConfig(input_size=256, sparse=True, minibatch_size=32,
+                       show_time_per_sample=0)
+    lr, _ = _train_local(cfg, 10)
+    p = str(tmp_path / "model.bin")
+    lr.save_model(p)
+    lr2 = LogReg(LogRegConfig(input_size=256, sparse=True,
+                              show_time_per_sample=0))
+    lr2.load_model(p)
+    assert torch.equal(lr.model.weight, lr2.model.weight)
+
+
+def test_ps_model_single(env):
+    cfg = LogRegConfig(input_size=4096, minibatch_size=64, use_ps=True,
+                       sync_frequency=4, learning_rate=0.05,
+                       learning_rate_coef=1e6, show_time_per_sample=0)
+    _, acc = _train_local(cfg)
+    assert acc > 0.75, acc
+
+
+def test_ps_model_adagrad_single(env):
+    cfg = LogRegConfig(input_size=4096, minibatch_size=64, use_ps=True,
+                       updater_type="adagrad", sync_frequency=4,
+                       learning_rate=0.05, show_time_per_sample=0)
+    _, acc = _train_local(cfg)
+    assert acc > 0.7, acc
+
+
+def test_ps_model_ftrl_single(env):
+    cfg = LogRegConfig(input_size=4096, minibatch_size=64, use_ps=True,
+                       objective_type="ftrl", sync_frequency=4,
+                       show_time_per_sample=0)
+    _, acc = _train_local(cfg)
+    assert acc > 0.7, acc
+
+
+def _ps_dist(rank, world):
+    import multiverso_amd as mv
+    from multiverso_amd.apps.logreg import LogReg, LogRegConfig
+    from multiverso_amd.apps.logreg.reader import synthetic_batches
+    mv.init(sync=True)
+    cfg = LogRegConfig(input_size=2048, minibatch_size=32, use_ps=True,
+                       sync_frequency=2, learning_rate=0.05,
+                       learning_rate_coef=1e6, show_time_per_sample=0)
+    batches, _ = synthetic_batches(cfg.input_size, 40, cfg.minibatch_size,
+                                   nnz=16, seed=100 + rank)
+    lr = LogReg(cfg)
+    lr.train(iter(batches))
+    acc, _ = lr.test(iter(batches[:10]))
+    assert acc > 0.7, (rank, acc)
+    mv.shutdown()
+
+
+def test_ps_dist():
+    run_dist(_ps_dist, 2)
