@@ -22,8 +22,9 @@ namespace py = pybind11;
 
 namespace {
 
-std::unique_ptr<py::scoped_interpreter> g_interp;
-bool g_owns_interpreter = false;
+// Deliberately leaked: destroying the embedded interpreter at process
+// exit races with torch/HIP static teardown (observed SIGSEGV at exit).
+py::scoped_interpreter* g_interp = nullptr;
 
 struct Handle {
   py::object table;  // ArrayTable or MatrixTable
@@ -48,8 +49,7 @@ extern "C" {
 
 void MV_Init(int* argc, char* argv[]) {
   if (!Py_IsInitialized()) {
-    g_interp = std::make_unique<py::scoped_interpreter>();
-    g_owns_interpreter = true;
+    g_interp = new py::scoped_interpreter();
   }
   py::gil_scoped_acquire gil;
   py::list args;
