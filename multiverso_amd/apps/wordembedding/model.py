@@ -56,6 +56,10 @@ class WordEmbeddingOption:
     # Duplicate-heavy blocks are chunked into sequential launches so later
     # occurrences of a pair see earlier updates — bounded staleness.
     max_groups_per_launch: int = 1 << 20
+    # Row updates race hogwild-style by default (the reference's own
+    # unsynchronized OpenMP semantics; 11x faster than atomics on
+    # MI355X). Set True for exact atomic accumulation.
+    atomic_updates: bool = False
 
 
 class WordEmbedding:
@@ -300,7 +304,8 @@ class WordEmbedding:
                               out_local[o0:o1].contiguous(),
                               out_label[o0:o1].contiguous(),
                               (out_off[g0:g1 + 1] - o0).contiguous(),
-                              self.learning_rate, self.opt.use_adagrad)
+                              self.learning_rate, self.opt.use_adagrad,
+                              self.opt.atomic_updates)
         else:
             _w2v_train_torch(in_buf, out_buf, in_gsq, out_gsq, in_local,
                              in_off, out_local, out_label, out_off,

@@ -21,7 +21,7 @@ void mv_launch_row_scatter_add(float*, const float*, const int64_t*, float,
                                int64_t, int64_t, hipStream_t);
 void mv_launch_w2v(float*, float*, float*, float*, const int64_t*, const int*,
                    const int64_t*, const float*, const int*, float, int64_t,
-                   int64_t, int, hipStream_t);
+                   int64_t, int, int, hipStream_t);
 void mv_launch_row_scatter_adagrad(float*, float*, const float*,
                                    const int64_t*, float, float, float,
                                    int64_t, int64_t, hipStream_t);
@@ -127,7 +127,8 @@ void w2v_train(torch::Tensor in_emb, torch::Tensor out_emb,
                torch::Tensor in_gsq, torch::Tensor out_gsq,
                torch::Tensor in_idx, torch::Tensor in_off,
                torch::Tensor out_idx, torch::Tensor out_label,
-               torch::Tensor out_off, double lr, bool use_adagrad) {
+               torch::Tensor out_off, double lr, bool use_adagrad,
+               bool use_atomic) {
   check_f32(in_emb, "in_emb"); check_f32(out_emb, "out_emb");
   check_f32(out_label, "out_label");
   TORCH_CHECK(in_emb.dim() == 2 && out_emb.dim() == 2, "emb must be 2-D");
@@ -151,7 +152,7 @@ void w2v_train(torch::Tensor in_emb, torch::Tensor out_emb,
                 in_idx.data_ptr<int64_t>(), in_off.data_ptr<int>(),
                 out_idx.data_ptr<int64_t>(), out_label.data_ptr<float>(),
                 out_off.data_ptr<int>(), (float)lr, G, in_emb.size(1),
-                use_adagrad ? 1 : 0, cur_stream());
+                use_adagrad ? 1 : 0, use_atomic ? 1 : 0, cur_stream());
 }
 
 }  // namespace

@@ -40,7 +40,7 @@ def test_w2v_kernel_vs_reference(dim, nin):
     d = lambda t: t.cuda()
     gin, gout = d(in_buf), d(out_buf)
     hip.w2v_train(gin, gout, gin, gout, d(in_idx), d(in_off), d(out_idx),
-                  d(out_label), d(out_off), lr, False)
+                  d(out_label), d(out_off), lr, False, False)
     torch.cuda.synchronize()
     assert torch.allclose(gin.cpu(), ref_in, rtol=1e-4, atol=1e-5), \
         (gin.cpu() - ref_in).abs().max()
@@ -64,7 +64,7 @@ def test_w2v_kernel_adagrad_vs_reference():
     d = lambda t: t.cuda()
     g = [d(in_buf), d(out_buf), d(in_gsq), d(out_gsq)]
     hip.w2v_train(g[0], g[1], g[2], g[3], d(in_idx), d(in_off), d(out_idx),
-                  d(out_label), d(out_off), lr, True)
+                  d(out_label), d(out_off), lr, True, False)
     torch.cuda.synchronize()
     for got, want in zip(g, ref):
         assert torch.allclose(got.cpu(), want, rtol=1e-3, atol=1e-5), \

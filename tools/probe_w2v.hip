@@ -1,0 +1,234 @@
+// A/B probe for the k_w2v kernel bottleneck on gfx950.
+// Variants: 0 = current (scalar f32 atomics), 1 = plain stores (racy
+// ceiling — perf bound only), 2 = packed v2f32 atomics (half the atomic
+// transactions), 3 = no in/out updates at all (pure read+dot ceiling).
+// Prints ms per launch for each variant on a Zipf-like group mix.
+//
+// build: hipcc --offload-arch=gfx950 -O3 -std=c++17 tools/probe_w2v.hip -o gpurun_out/probe_w2v
+// run:   ./gpurun_out/probe_w2v [G] [V] [dim] [neg]
+
+#include <hip/hip_runtime.h>
+#include <cstdio>
+#include <cstdlib>
+#include <cmath>
+#include <vector>
+
+#define BLOCK 256
+#define MAX_GRID 2048
+
+typedef float v2f __attribute__((ext_vector_type(2)));
+
+// VARIANT 2: float2 loads/atomics — lane owns 2 consecutive floats
+// (8B/lane, 512B/wave-instruction); separate body below.
+template <int DPL2>
+__global__ void k_w2v_probe_f2(float* __restrict__ in_emb,
+                               float* __restrict__ out_emb,
+                               const long* __restrict__ in_idx,
+                               const int* __restrict__ in_off,
+                               const long* __restrict__ out_idx,
+                               const float* __restrict__ out_label,
+                               const int* __restrict__ out_off,
+                               float lr, int G, int dim) {
+  int wid = (int)((blockIdx.x * (long)blockDim.x + threadIdx.x) >> 6);
+  int lane = threadIdx.x & 63;
+  int nwaves = (int)((gridDim.x * (long)blockDim.x) >> 6);
+  for (int g = wid; g < G; g += nwaves) {
+    v2f h[DPL2], err[DPL2];
+#pragma unroll
+    for (int d = 0; d < DPL2; ++d) { h[d] = {0.f, 0.f}; err[d] = {0.f, 0.f}; }
+    int ib = in_off[g], ie = in_off[g + 1];
+    for (int i = ib; i < ie; ++i) {
+      const float* row = in_emb + in_idx[i] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL2; ++d) {
+        int c = lane * 2 + 128 * d;
+        if (c + 1 < dim) { v2f v = *(const v2f*)(row + c); h[d] += v; }
+        else if (c < dim) { h[d].x += row[c]; }
+      }
+    }
+    int ob = out_off[g], oe = out_off[g + 1];
+    for (int o = ob; o < oe; ++o) {
+      float* w = out_emb + out_idx[o] * dim;
+      v2f wv[DPL2];
+      float f = 0.f;
+#pragma unroll
+      for (int d = 0; d < DPL2; ++d) {
+        int c = lane * 2 + 128 * d;
+        wv[d] = {0.f, 0.f};
+        if (c + 1 < dim) wv[d] = *(const v2f*)(w + c);
+        else if (c < dim) wv[d].x = w[c];
+        f += h[d].x * wv[d].x + h[d].y * wv[d].y;
+      }
+#pragma unroll
+      for (int s = 32; s; s >>= 1) f += __shfl_xor(f, s, 64);
+      f = 1.f / (1.f + expf(-f));
+      float e = out_label[o] - f;
+#pragma unroll
+      for (int d = 0; d < DPL2; ++d) {
+        int c = lane * 2 + 128 * d;
+        err[d] += e * wv[d];
+        if (c + 1 < dim) {
+          atomicAdd(&w[c], e * lr * h[d].x);
+          atomicAdd(&w[c + 1], e * lr * h[d].y);
+        } else if (c < dim) {
+          atomicAdd(&w[c], e * lr * h[d].x);
+        }
+      }
+    }
+    for (int i = ib; i < ie; ++i) {
+      float* row = in_emb + in_idx[i] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL2; ++d) {
+        int c = lane * 2 + 128 * d;
+        if (c + 1 < dim) {
+          atomicAdd(&row[c], lr * err[d].x);
+          atomicAdd(&row[c + 1], lr * err[d].y);
+        } else if (c < dim) {
+          atomicAdd(&row[c], lr * err[d].x);
+        }
+      }
+    }
+  }
+}
+
+template <int DPL, int VARIANT>
+__global__ void k_w2v_probe(float* __restrict__ in_emb,
+                            float* __restrict__ out_emb,
+                            const long* __restrict__ in_idx,
+                            const int* __restrict__ in_off,
+                            const long* __restrict__ out_idx,
+                            const float* __restrict__ out_label,
+                            const int* __restrict__ out_off,
+                            float lr, int G, int dim) {
+  int wid = (int)((blockIdx.x * (long)blockDim.x + threadIdx.x) >> 6);
+  int lane = threadIdx.x & 63;
+  int nwaves = (int)((gridDim.x * (long)blockDim.x) >> 6);
+  for (int g = wid; g < G; g += nwaves) {
+    float h[DPL], err[DPL];
+#pragma unroll
+    for (int d = 0; d < DPL; ++d) { h[d] = 0.f; err[d] = 0.f; }
+    int ib = in_off[g], ie = in_off[g + 1];
+    for (int i = ib; i < ie; ++i) {
+      const float* row = in_emb + in_idx[i] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) h[d] += row[c];
+      }
+    }
+    int ob = out_off[g], oe = out_off[g + 1];
+    for (int o = ob; o < oe; ++o) {
+      float* w = out_emb + out_idx[o] * dim;
+      float wv[DPL];
+      float f = 0.f;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        wv[d] = (c < dim) ? w[c] : 0.f;
+        f += h[d] * wv[d];
+      }
+#pragma unroll
+      for (int s = 32; s; s >>= 1) f += __shfl_xor(f, s, 64);
+      f = 1.f / (1.f + expf(-f));
+      float e = out_label[o] - f;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) {
+          err[d] += e * wv[d];
+          if (VARIANT == 0) {
+            atomicAdd(&w[c], e * lr * h[d]);
+          } else if (VARIANT == 1) {
+            w[c] = wv[d] + e * lr * h[d];
+          } else if (VARIANT == 3) {
+            // no update
+          }
+        }
+      }
+    }
+    for (int i = ib; i < ie; ++i) {
+      float* row = in_emb + in_idx[i] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) {
+          if (VARIANT == 0) atomicAdd(&row[c], lr * err[d]);
+          else if (VARIANT == 1) row[c] += lr * err[d];
+        }
+      }
+    }
+  }
+}
+
+int main(int argc, char** argv) {
+  int G = argc > 1 ? atoi(argv[1]) : 1 << 21;
+  int V = argc > 2 ? atoi(argv[2]) : 1000000;  // vocab
+  int dim = argc > 3 ? atoi(argv[3]) : 200;
+  int neg = argc > 4 ? atoi(argv[4]) : 5;
+  int K = 1 + neg;
+
+  float *in_emb, *out_emb, *labels;
+  long *in_idx, *out_idx;
+  int *in_off, *out_off;
+  hipMalloc(&in_emb, (size_t)V * dim * 4);
+  hipMalloc(&out_emb, (size_t)V * dim * 4);
+  hipMalloc(&in_idx, (size_t)G * 8);
+  hipMalloc(&out_idx, (size_t)G * K * 8);
+  hipMalloc(&labels, (size_t)G * K * 4);
+  hipMalloc(&in_off, (size_t)(G + 1) * 4);
+  hipMalloc(&out_off, (size_t)(G + 1) * 4);
+
+  // host-side Zipf-ish ids
+  std::vector<long> h_in(G), h_out((size_t)G * K);
+  std::vector<int> h_ioff(G + 1), h_ooff(G + 1);
+  std::vector<float> h_lab((size_t)G * K);
+  srand(7);
+  auto zipf = [&](void) -> long {
+    double u = (rand() + 1.0) / (RAND_MAX + 2.0);
+    long id = (long)(exp(u * log((double)V))) - 1;
+    return id < 0 ? 0 : (id >= V ? V - 1 : id);
+  };
+  for (int g = 0; g < G; ++g) {
+    h_in[g] = zipf();
+    h_ioff[g] = g;
+    h_ooff[g] = g * K;
+    for (int k = 0; k < K; ++k) {
+      h_out[(size_t)g * K + k] = zipf();
+      h_lab[(size_t)g * K + k] = k == 0 ? 1.f : 0.f;
+    }
+  }
+  h_ioff[G] = G; h_ooff[G] = G * K;
+  hipMemcpy(in_idx, h_in.data(), G * 8, hipMemcpyHostToDevice);
+  hipMemcpy(out_idx, h_out.data(), (size_t)G * K * 8, hipMemcpyHostToDevice);
+  hipMemcpy(labels, h_lab.data(), (size_t)G * K * 4, hipMemcpyHostToDevice);
+  hipMemcpy(in_off, h_ioff.data(), (G + 1) * 4, hipMemcpyHostToDevice);
+  hipMemcpy(out_off, h_ooff.data(), (G + 1) * 4, hipMemcpyHostToDevice);
+  hipMemset(in_emb, 0, (size_t)V * dim * 4);
+  hipMemset(out_emb, 0, (size_t)V * dim * 4);
+
+  int grid = (int)std::min((long)MAX_GRID, ((long)G * 64 + BLOCK - 1) / BLOCK);
+
+  auto bench = [&](auto kern, const char* name) {
+    // warmup
+    kern<<<grid, BLOCK>>>(in_emb, out_emb, in_idx, in_off, out_idx, labels,
+                          out_off, 0.025f, G, dim);
+    hipDeviceSynchronize();
+    hipEvent_t a, b;
+    hipEventCreate(&a); hipEventCreate(&b);
+    hipEventRecord(a);
+    for (int r = 0; r < 3; ++r)
+      kern<<<grid, BLOCK>>>(in_emb, out_emb, in_idx, in_off, out_idx, labels,
+                            out_off, 0.025f, G, dim);
+    hipEventRecord(b);
+    hipEventSynchronize(b);
+    float ms;
+    hipEventElapsedTime(&ms, a, b);
+    printf("%s: %.3f ms/launch (G=%d)\n", name, ms / 3, G);
+  };
+
+  bench(k_w2v_probe<4, 0>, "V0 scalar-atomic");
+  bench(k_w2v_probe<4, 1>, "V1 plain-store  ");
+  bench(k_w2v_probe_f2<2>, "V2 float2-loads ");
+  bench(k_w2v_probe<4, 3>, "V3 read-only    ");
+  return 0;
+}
