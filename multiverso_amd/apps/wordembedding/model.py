@@ -249,11 +249,28 @@ class WordEmbedding:
             words, sent_ids = words[m], sent_ids[m]
         in_idx, in_off, out_idx, out_label, out_off = \
             self.build_groups(words, sent_ids)
+        if in_idx.numel() == 0 or out_idx.numel() == 0:
+            return int(words.numel())
+
+        if mv.size() == 1:
+            # Single-rank fast path: every row is local, and
+            # pull → train → push(trained − pulled) is identical to
+            # training the HBM-resident shard in place (P=1). Skips two
+            # unique-sorts, the gathers/scatters and the delta pass.
+            igq = ogq = None
+            if opt.use_adagrad:
+                igq = self.input_gsq_table.shard
+                ogq = self.output_gsq_table.shard
+            self._train_kernel(self.input_table.shard,
+                               self.output_table.shard, igq, ogq,
+                               in_idx, in_off, out_idx,
+                               out_label.float(), out_off)
+            nwords = int(words.numel())
+            self._update_lr(nwords)
+            return nwords
 
         uin, in_local = torch.unique(in_idx, return_inverse=True)
         uout, out_local = torch.unique(out_idx, return_inverse=True)
-        if uin.numel() == 0 or uout.numel() == 0:
-            return int(words.numel())
 
         # pull touched rows (RequestParameter)
         in_buf = self.input_table.get_rows(uin).contiguous()

@@ -162,7 +162,6 @@ def all_to_all_rows(row_ids: torch.Tensor, values: Optional[torch.Tensor],
     """
     device = row_ids.device
     if not _dist():
-        order = torch.argsort(row_ids, stable=True)
         return row_ids, values, [row_ids.numel()], torch.arange(
             row_ids.numel(), device=device)
     n = dist.get_world_size()
