@@ -1,0 +1,70 @@
+// A/B probe for the elementwise SGD updater (K2) roofline on gfx950:
+// variants x grid sizes; traffic = read delta + read/write data = 12 B
+// per element. Target: ~6.3 TB/s achievable HBM.
+// build: hipcc --offload-arch=gfx950 -O3 tools/probe_sgd.hip -o probe_sgd
+#include <hip/hip_runtime.h>
+#include <cstdio>
+
+#define BLOCK 256
+typedef float v4f __attribute__((ext_vector_type(4)));
+
+__global__ void sgd_f4(float4* __restrict__ d, const float4* __restrict__ g, long n4) {
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    float4 a = d[i], b = g[i];
+    a.x -= b.x; a.y -= b.y; a.z -= b.z; a.w -= b.w;
+    d[i] = a;
+  }
+}
+
+__global__ void sgd_f4_nt(float4* __restrict__ d, const float4* __restrict__ g, long n4) {
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    float4 a = d[i];
+    v4f b = __builtin_nontemporal_load((const v4f*)&g[i]);
+    a.x -= b[0]; a.y -= b[1]; a.z -= b[2]; a.w -= b[3];
+    d[i] = a;
+  }
+}
+
+__global__ void sgd_f4_ntboth(float4* __restrict__ d, const float4* __restrict__ g, long n4) {
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    v4f a = __builtin_nontemporal_load((v4f*)&d[i]);
+    v4f b = __builtin_nontemporal_load((const v4f*)&g[i]);
+    a -= b;
+    __builtin_nontemporal_store(a, (v4f*)&d[i]);
+  }
+}
+
+int main() {
+  long n = 128L * 1000 * 1000;  // 1e6x128
+  long n4 = n / 4;
+  float *d, *g;
+  (void)hipMalloc(&d, n * 4);
+  (void)hipMalloc(&g, n * 4);
+  (void)hipMemset(d, 0, n * 4);
+  (void)hipMemset(g, 0, n * 4);
+  auto bench = [&](auto kern, int grid, const char* name) {
+    kern<<<grid, BLOCK>>>((float4*)d, (const float4*)g, n4);
+    (void)hipDeviceSynchronize();
+    hipEvent_t a, b;
+    (void)hipEventCreate(&a); (void)hipEventCreate(&b);
+    (void)hipEventRecord(a);
+    for (int r = 0; r < 10; ++r)
+      kern<<<grid, BLOCK>>>((float4*)d, (const float4*)g, n4);
+    (void)hipEventRecord(b);
+    (void)hipEventSynchronize(b);
+    float ms;
+    (void)hipEventElapsedTime(&ms, a, b);
+    ms /= 10;
+    printf("%s grid=%5d: %.3f ms  %.2f TB/s\n", name, grid, ms,
+           n * 12.0 / (ms * 1e-3) / 1e12);
+  };
+  for (int grid : {1024, 2048, 4096, 8192, 16384}) {
+    bench(sgd_f4, grid, "plain  ");
+    bench(sgd_f4_nt, grid, "nt-load");
+    bench(sgd_f4_ntboth, grid, "nt-both");
+  }
+  return 0;
+}
