@@ -95,7 +95,14 @@ def allgather_shards(out_flat: torch.Tensor, shard_flat: torch.Tensor,
     """
     if not _dist():
         off, cnt = spec.range_of(0)
-        out_flat[off * unit:(off + cnt) * unit].copy_(shard_flat)
+        dst = out_flat[off * unit:(off + cnt) * unit]
+        if (dst.is_cuda and dst.dtype == torch.float32
+                and dst.is_contiguous() and shard_flat.is_contiguous()):
+            from . import ops
+            m = ops.module(required=True)
+            m.nt_copy(dst, shard_flat)   # K7: NT streaming copy (5.8 TB/s)
+        else:
+            dst.copy_(shard_flat)
         return _NOOP
     if spec.even and out_flat.is_contiguous():
         work = dist.all_gather_into_tensor(out_flat, shard_flat, async_op=async_op)

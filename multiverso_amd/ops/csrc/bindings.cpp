@@ -10,6 +10,7 @@
 #include <cstdint>
 
 extern "C" {
+void mv_launch_copy(float*, const float*, int64_t, hipStream_t);
 void mv_launch_add(float*, const float*, int64_t, hipStream_t);
 void mv_launch_sgd(float*, const float*, int64_t, hipStream_t);
 void mv_launch_momentum(float*, float*, const float*, float, int64_t, hipStream_t);
@@ -37,6 +38,13 @@ void check_f32(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
   TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+}
+
+void nt_copy(torch::Tensor dst, torch::Tensor src) {
+  check_f32(dst, "dst"); check_f32(src, "src");
+  TORCH_CHECK(dst.numel() == src.numel(), "size mismatch");
+  mv_launch_copy(dst.data_ptr<float>(), src.data_ptr<float>(), dst.numel(),
+                 cur_stream());
 }
 
 void add_inplace(torch::Tensor data, torch::Tensor delta) {
@@ -159,6 +167,7 @@ void w2v_train(torch::Tensor in_emb, torch::Tensor out_emb,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_inplace", &add_inplace, "K1: data += delta (fp32, fused)");
+  m.def("nt_copy", &nt_copy, "K7: non-temporal streaming copy");
   m.def("sgd_update", &sgd_update, "K2: data -= delta");
   m.def("momentum_update", &momentum_update, "K3: fused momentum update");
   m.def("adagrad_update", &adagrad_update, "K4: fused adagrad update");
