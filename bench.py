@@ -94,8 +94,9 @@ def main():
     # MAX over ranks
     t = torch.tensor([elapsed], dtype=torch.float64)
     if dist.is_initialized() and n > 1:
-        dist.all_reduce(t.to(device) if mv.Zoo.get().backend == "nccl" else t,
-                        op=dist.ReduceOp.MAX)
+        if mv.Zoo.get().backend == "nccl":
+            t = t.to(device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed = float(t[0])
 
     ms_per_step = elapsed / args.steps * 1e3

@@ -62,8 +62,9 @@ def run_bench(args):
 
     t = torch.tensor([elapsed], dtype=torch.float64)
     if dist.is_initialized() and n > 1:
-        dist.all_reduce(t.to(device) if mv.Zoo.get().backend == "nccl" else t,
-                        op=dist.ReduceOp.MAX)
+        if mv.Zoo.get().backend == "nccl":
+            t = t.to(device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed = float(t[0])
     words_per_sec = n * total_words / elapsed
 
