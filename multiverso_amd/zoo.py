@@ -163,11 +163,31 @@ class Zoo:
             else:
                 dist.barrier()
 
-    def aggregate(self, tensor: torch.Tensor) -> torch.Tensor:
+    def aggregate(self, tensor: torch.Tensor,
+                  bucket_mb: Optional[int] = None) -> torch.Tensor:
         """In-place sum-allreduce — MV_Aggregate (src/multiverso.cpp:53-56,
-        net.cpp:27-41). On GPU this is one rcclAllReduce over xGMI."""
-        if dist.is_initialized():
+        net.cpp:27-41). On GPU this is rcclAllReduce over xGMI; payloads
+        larger than ``bucket_mb`` (flag ``bucket_mb``) are issued as a
+        pipeline of async bucket all-reduces — the rebuild's tunable for
+        the reference's 4096-byte small-message switch
+        (allreduce_engine.cpp:35), sized for 7 xGMI links instead."""
+        if not dist.is_initialized():
+            return tensor
+        if bucket_mb is None:
+            bucket_mb = int(get_flag("bucket_mb"))
+        nbytes = tensor.numel() * tensor.element_size()
+        if bucket_mb <= 0 or nbytes <= bucket_mb * (1 << 20):
             dist.all_reduce(tensor, op=dist.ReduceOp.SUM)
+            return tensor
+        flat = tensor.view(-1)
+        step = bucket_mb * (1 << 20) // tensor.element_size()
+        works = []
+        for off in range(0, flat.numel(), step):
+            works.append(dist.all_reduce(flat[off:off + step],
+                                         op=dist.ReduceOp.SUM,
+                                         async_op=True))
+        for w in works:
+            w.wait()
         return tensor
 
     @property
