@@ -87,9 +87,11 @@ def size() -> int:
     return Zoo.get().size
 
 
-def aggregate(tensor: torch.Tensor) -> torch.Tensor:
-    """MV_Aggregate — in-place sum-allreduce (model-average mode)."""
-    return Zoo.get().aggregate(tensor)
+def aggregate(tensor: torch.Tensor,
+              bucket_mb: Optional[int] = None) -> torch.Tensor:
+    """MV_Aggregate — in-place sum-allreduce (model-average mode),
+    bucket-pipelined above the ``bucket_mb`` threshold."""
+    return Zoo.get().aggregate(tensor, bucket_mb=bucket_mb)
 
 
 # ---------------------------------------------------------------------------
