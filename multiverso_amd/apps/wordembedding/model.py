@@ -89,6 +89,7 @@ class WordEmbedding:
             self.huffman.build_from_term_frequency(counts)
             self._build_hs_tensors()
         self.learning_rate = option.init_learning_rate
+        self._block_pool: Optional[torch.Tensor] = None
         self.word_count_local = 0      # words since last global sync
         self.word_count_actual = 0     # global processed words
         self.gen = torch.Generator(device=self.device)
@@ -187,14 +188,24 @@ class WordEmbedding:
         return in_idx, in_off, out_idx, out_label, out_off
 
     def _outputs_for(self, centers: torch.Tensor):
-        """Output node/label lists per group for NS and/or HS."""
+        """Output node/label lists per group for NS and/or HS. Negatives
+        draw from the per-block pool (reference PrepareData,
+        wordembedding.cpp:190-209: negative_num x |unique inputs| samples
+        per block) — this bounds the block's touched output rows."""
         opt = self.opt
         device = self.device
         g = centers.numel()
         idx_parts, label_parts, len_parts = [], [], []
         if opt.negative_num > 0:
-            negs = self.sampler.negative_sampling((g, opt.negative_num),
-                                                  generator=self.gen)
+            pool = self._block_pool
+            if pool is not None and pool.numel() > 0:
+                pick = torch.randint(0, pool.numel(),
+                                     (g, opt.negative_num), device=device,
+                                     generator=self.gen)
+                negs = pool[pick]
+            else:
+                negs = self.sampler.negative_sampling(
+                    (g, opt.negative_num), generator=self.gen)
             outs = torch.cat([centers.unsqueeze(1), negs], dim=1)
             labels = torch.zeros(g, 1 + opt.negative_num, device=device)
             labels[:, 0] = 1.0
@@ -247,6 +258,11 @@ class WordEmbedding:
         if opt.sample > 0:
             m = self.sampler.keep_mask(words, opt.sample, self.gen)
             words, sent_ids = words[m], sent_ids[m]
+        uwords = torch.unique(words)
+        self._block_pool = None
+        if opt.negative_num > 0:
+            self._block_pool = self.sampler.negative_sampling(
+                (opt.negative_num * uwords.numel(),), generator=self.gen)
         in_idx, in_off, out_idx, out_label, out_off = \
             self.build_groups(words, sent_ids)
         if in_idx.numel() == 0 or out_idx.numel() == 0:
@@ -269,8 +285,21 @@ class WordEmbedding:
             self._update_lr(nwords)
             return nwords
 
-        uin, in_local = torch.unique(in_idx, return_inverse=True)
-        uout, out_local = torch.unique(out_idx, return_inverse=True)
+        # Touched-row candidate sets are known up front (block vocabulary
+        # + negative pool + HS path nodes), so the 10M+-element
+        # unique-sorts collapse to searchsorted over small sorted sets.
+        uin = uwords  # inputs are always block words
+        in_local = torch.searchsorted(uin, in_idx)
+        cand = [uwords] if (opt.negative_num > 0 or not opt.hs) else []
+        if self._block_pool is not None:
+            cand.append(self._block_pool)
+        if opt.hs:
+            starts = self._hs_off[uwords]
+            lens = self._hs_off[uwords + 1] - starts
+            cand.append(self._hs_point[
+                starts.repeat_interleave(lens) + _segment_arange(lens)])
+        uout = torch.unique(torch.cat(cand))
+        out_local = torch.searchsorted(uout, out_idx)
 
         # pull touched rows (RequestParameter)
         in_buf = self.input_table.get_rows(uin).contiguous()

@@ -142,3 +142,35 @@ def test_save_embedding(env, tmp_path):
     assert len(lines) == 21
     assert lines[1].split()[0] == "w0"
     assert len(lines[1].split()) == 9
+
+
+def _w2v_dist(rank, world):
+    import torch
+    import multiverso_amd as mv
+    from multiverso_amd.apps.wordembedding.model import (WordEmbedding,
+                                                         WordEmbeddingOption)
+    mv.init(sync=True)
+    torch.manual_seed(rank)
+    opt = WordEmbeddingOption(embedding_size=16, window=1, negative_num=3,
+                              init_learning_rate=0.1,
+                              total_words=10_000_000, seed=5)
+    model = WordEmbedding(opt, [100] * 12)
+    words = torch.stack([torch.arange(0, 12, 2).repeat(30),
+                         torch.arange(1, 12, 2).repeat(30)], dim=1).view(-1)
+    sids = torch.arange(words.numel()) // 10
+    for _ in range(6):
+        model.train_block(words, sids)       # multi-rank pull/push path
+        model.sync_word_count()
+    assert model.word_count_actual == 6 * words.numel() * world
+    inp = model.input_table.get()
+    out = model.output_table.get()
+    evens = torch.arange(0, 12, 2)
+    pos = torch.sigmoid((inp[evens] * out[evens + 1]).sum(1)).mean()
+    wrong = torch.sigmoid((inp[evens] * out[evens.roll(1) + 1]).sum(1)).mean()
+    assert float(pos) - float(wrong) > 0.2, (rank, float(pos), float(wrong))
+    mv.shutdown()
+
+
+def test_w2v_dist_two_ranks():
+    from conftest import run_dist
+    run_dist(_w2v_dist, 2)
