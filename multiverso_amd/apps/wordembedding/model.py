@@ -209,10 +209,13 @@ class WordEmbedding:
             outs = torch.cat([centers.unsqueeze(1), negs], dim=1)
             labels = torch.zeros(g, 1 + opt.negative_num, device=device)
             labels[:, 0] = 1.0
-            idx_parts.append(outs)
+            # a negative that collides with the positive target is dropped,
+            # not trained with label 0 (wordembedding.cpp:279 `continue`)
+            keep = torch.ones_like(outs, dtype=torch.bool)
+            keep[:, 1:] = negs != centers.unsqueeze(1)
+            idx_parts.append((outs, keep))
             label_parts.append(labels)
-            len_parts.append(torch.full((g,), 1 + opt.negative_num,
-                                        dtype=torch.int64, device=device))
+            len_parts.append(keep.sum(1))
         if opt.hs:
             starts = self._hs_off[centers]
             lens = self._hs_off[centers + 1] - starts
@@ -222,27 +225,27 @@ class WordEmbedding:
             hs_lab = self._hs_label[flat_pos]
             if idx_parts:
                 # NS + HS together: interleave per group
-                ns_idx, ns_lab = idx_parts[0], label_parts[0]
-                k = ns_idx.size(1)
-                out_lens = lens + k
+                (ns_idx, ns_keep), ns_lab = idx_parts[0], label_parts[0]
+                ns_lens = len_parts[0]
+                out_lens = lens + ns_lens
                 off = _lens_to_off(out_lens)
                 total = int(out_lens.sum())
                 out_idx = torch.empty(total, dtype=torch.int64, device=device)
                 out_lab = torch.empty(total, device=device)
-                ns_pos = (off[:-1].repeat_interleave(k)
-                          + _segment_arange(torch.full_like(lens, k)))
-                out_idx[ns_pos] = ns_idx.reshape(-1)
-                out_lab[ns_pos] = ns_lab.reshape(-1)
-                hs_pos = ((off[:-1] + k).repeat_interleave(lens)
+                ns_pos = (off[:-1].repeat_interleave(ns_lens)
+                          + _segment_arange(ns_lens))
+                out_idx[ns_pos] = ns_idx[ns_keep]
+                out_lab[ns_pos] = ns_lab[ns_keep]
+                hs_pos = ((off[:-1] + ns_lens).repeat_interleave(lens)
                           + _segment_arange(lens))
                 out_idx[hs_pos] = hs_idx
                 out_lab[hs_pos] = hs_lab
                 return out_idx, out_lab, off.to(torch.int32)
             off = _lens_to_off(lens)
             return hs_idx, hs_lab, off.to(torch.int32)
-        ns_idx, ns_lab = idx_parts[0], label_parts[0]
+        (ns_idx, ns_keep), ns_lab = idx_parts[0], label_parts[0]
         off = _lens_to_off(len_parts[0])
-        return ns_idx.reshape(-1), ns_lab.reshape(-1), off.to(torch.int32)
+        return ns_idx[ns_keep], ns_lab[ns_keep], off.to(torch.int32)
 
     # ------------------------------------------------------------------
     # Block training
@@ -261,8 +264,10 @@ class WordEmbedding:
         uwords = torch.unique(words)
         self._block_pool = None
         if opt.negative_num > 0:
-            self._block_pool = self.sampler.negative_sampling(
-                (opt.negative_num * uwords.numel(),), generator=self.gen)
+            # deduped, like the reference's negativesample_pools std::set
+            # (wordembedding.cpp:208)
+            self._block_pool = torch.unique(self.sampler.negative_sampling(
+                (opt.negative_num * uwords.numel(),), generator=self.gen))
         in_idx, in_off, out_idx, out_label, out_off = \
             self.build_groups(words, sent_ids)
         if in_idx.numel() == 0 or out_idx.numel() == 0:

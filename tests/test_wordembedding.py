@@ -74,11 +74,21 @@ def test_group_building_skipgram(env):
     g = in_off.numel() - 1
     assert g > 0
     assert out_off.numel() - 1 == g
-    assert int(out_off[-1]) == g * 4  # 1 pos + 3 negs per group
-    lab = out_label.view(g, 4)
-    assert torch.all(lab[:, 0] == 1) and torch.all(lab[:, 1:] == 0)
+    # 1 pos + up to 3 negs per group (negatives colliding with the target
+    # are dropped, wordembedding.cpp:279)
+    lens = (out_off[1:] - out_off[:-1]).long()
+    assert torch.all(lens >= 1) and torch.all(lens <= 4)
+    assert int(lens.sum()) == int(out_off[-1])
+    first = out_off[:-1].long()
+    assert torch.all(out_label[first] == 1)  # positive leads each group
+    neg_mask = torch.ones(out_idx.numel(), dtype=torch.bool)
+    neg_mask[first] = False
+    assert torch.all(out_label[neg_mask] == 0)
+    # negatives never equal their group's positive target
+    pos = out_idx[first]
+    gid = torch.repeat_interleave(torch.arange(g), lens)
+    assert torch.all(out_idx[neg_mask] != pos[gid[neg_mask]])
     # positives are within window distance of their input
-    pos = out_idx.view(g, 4)[:, 0]
     assert torch.all((pos - in_idx).abs() <= opt.window)
 
 
