@@ -52,10 +52,13 @@ class WordEmbeddingOption:
     unigram_table_size: int = 10_000_000
     seed: int = 1
     # Within a kernel launch, groups train hogwild-concurrently (row reads
-    # are pre-update, like the reference's unsynchronized OpenMP trainers).
-    # Duplicate-heavy blocks are chunked into sequential launches so later
-    # occurrences of a pair see earlier updates — bounded staleness.
-    max_groups_per_launch: int = 1 << 20
+    # are pre-update, like the reference's unsynchronized OpenMP trainers,
+    # which give no ordering guarantee within a block either). Setting this
+    # below a block's group count chunks the block into sequential launches
+    # so later occurrences of a pair see earlier updates — bounded
+    # staleness, STRICTER than reference semantics; useful for tiny
+    # duplicate-heavy corpora (see tests).
+    max_groups_per_launch: int = 1 << 23
     # Row updates race hogwild-style by default (the reference's own
     # unsynchronized OpenMP semantics; 11x faster than atomics on
     # MI355X). Set True for exact atomic accumulation.
@@ -122,28 +125,33 @@ class WordEmbedding:
             return self._build_cbow_groups(words, sent_ids)
         device = self.device
         n = words.numel()
-        centers_l, contexts_l = [], []
-        for o in range(1, opt.window + 1):
-            if n <= o:
-                break
-            left = torch.arange(0, n - o, device=device)
-            valid = sent_ids[left] == sent_ids[left + o]
-            # reduced-window: offset o survives with prob (window-o+1)/window
-            keep = torch.rand(left.numel(), device=device,
-                              generator=self.gen) < (opt.window - o + 1) / opt.window
-            sel = left[valid & keep]
-            if sel.numel() == 0:
-                continue
-            # center at i, context at i+o  AND  center at i+o, context at i
-            centers_l.append(words[sel]); contexts_l.append(words[sel + o])
-            centers_l.append(words[sel + o]); contexts_l.append(words[sel])
-        if not centers_l:
+        W = min(opt.window, n - 1)
+        if W < 1:
             e = torch.empty(0, dtype=torch.int64, device=device)
             z = torch.zeros(1, dtype=torch.int32, device=device)
             return e, z, e, e.float(), z
-
-        centers = torch.cat(centers_l)
-        contexts = torch.cat(contexts_l)
+        # all window offsets in one batched pass: pair (i, i+o) for
+        # o = 1..W, valid when in-range + same sentence + reduced-window
+        # keep (offset o survives with prob (window-o+1)/window)
+        offs = torch.arange(1, W + 1, device=device)            # (W,)
+        left = torch.arange(n, device=device).unsqueeze(1)      # (n,1)
+        right = left + offs                                     # (n,W)
+        inb = right < n
+        right_c = right.clamp(max=n - 1)
+        valid = inb & (sent_ids.unsqueeze(1) == sent_ids[right_c])
+        keep = torch.rand((n, W), device=device, generator=self.gen) \
+            < (opt.window - offs + 1).float() / opt.window
+        m = (valid & keep).reshape(-1)
+        li = left.expand(n, W).reshape(-1)[m]
+        ri = right.reshape(-1)[m]
+        if li.numel() == 0:
+            e = torch.empty(0, dtype=torch.int64, device=device)
+            z = torch.zeros(1, dtype=torch.int32, device=device)
+            return e, z, e, e.float(), z
+        # center at i, context at i+o  AND  center at i+o, context at i
+        wl, wr = words[li], words[ri]
+        centers = torch.cat([wl, wr])
+        contexts = torch.cat([wr, wl])
 
         # skip-gram: one group per pair, input = context word
         g = centers.numel()
@@ -157,25 +165,28 @@ class WordEmbedding:
         opt = self.opt
         device = self.device
         n = words.numel()
-        pos_c, ctx_w = [], []
-        for o in range(1, opt.window + 1):
-            if n <= o:
-                break
-            left = torch.arange(0, n - o, device=device)
-            valid = sent_ids[left] == sent_ids[left + o]
-            keep = torch.rand(left.numel(), device=device,
-                              generator=self.gen) < (opt.window - o + 1) / opt.window
-            sel = left[valid & keep]
-            if sel.numel() == 0:
-                continue
-            pos_c.append(sel); ctx_w.append(words[sel + o])        # center i
-            pos_c.append(sel + o); ctx_w.append(words[sel])        # center i+o
-        if not pos_c:
+        W = min(opt.window, n - 1)
+        if W < 1:
             e = torch.empty(0, dtype=torch.int64, device=device)
             z = torch.zeros(1, dtype=torch.int32, device=device)
             return e, z, e, e.float(), z
-        pos = torch.cat(pos_c)
-        ctx = torch.cat(ctx_w)
+        offs = torch.arange(1, W + 1, device=device)
+        left = torch.arange(n, device=device).unsqueeze(1)
+        right = left + offs
+        inb = right < n
+        right_c = right.clamp(max=n - 1)
+        valid = inb & (sent_ids.unsqueeze(1) == sent_ids[right_c])
+        keep = torch.rand((n, W), device=device, generator=self.gen) \
+            < (opt.window - offs + 1).float() / opt.window
+        m = (valid & keep).reshape(-1)
+        li = left.expand(n, W).reshape(-1)[m]
+        ri = right.reshape(-1)[m]
+        if li.numel() == 0:
+            e = torch.empty(0, dtype=torch.int64, device=device)
+            z = torch.zeros(1, dtype=torch.int32, device=device)
+            return e, z, e, e.float(), z
+        pos = torch.cat([li, ri])      # center positions
+        ctx = torch.cat([words[ri], words[li]])
         order = torch.argsort(pos, stable=True)
         pos, ctx = pos[order], ctx[order]
         upos, counts = torch.unique_consecutive(pos, return_counts=True)
@@ -345,10 +356,23 @@ class WordEmbedding:
             out_off = out_off.to(torch.int32)
             g_total = in_off.numel() - 1
             step = self.opt.max_groups_per_launch
-            for g0 in range(0, g_total, step):
-                g1 = min(g0 + step, g_total)
-                i0, i1 = int(in_off[g0]), int(in_off[g1])
-                o0, o1 = int(out_off[g0]), int(out_off[g1])
+            if g_total <= step:
+                # single launch: no slicing, no host sync
+                hip.w2v_train(in_buf, out_buf, igq, ogq,
+                              in_local, in_off, out_local, out_label,
+                              out_off, self.learning_rate,
+                              self.opt.use_adagrad,
+                              self.opt.atomic_updates)
+                return
+            # one host transfer for every chunk boundary (int(in_off[g])
+            # per chunk would sync the stream 4x per chunk)
+            gs = list(range(0, g_total, step)) + [g_total]
+            ibs = in_off[gs].tolist()
+            obs = out_off[gs].tolist()
+            for k in range(len(gs) - 1):
+                g0, g1 = gs[k], gs[k + 1]
+                i0, i1 = ibs[k], ibs[k + 1]
+                o0, o1 = obs[k], obs[k + 1]
                 hip.w2v_train(in_buf, out_buf, igq, ogq,
                               in_local[i0:i1].contiguous(),
                               (in_off[g0:g1 + 1] - i0).contiguous(),

@@ -160,6 +160,93 @@ __global__ void k_w2v_probe(float* __restrict__ in_emb,
   }
 }
 
+// VARIANT 4: plain stores + tile-of-8 output-row prefetch. All output rows
+// of a tile are loaded before any dot/update, so the per-group row fetches
+// overlap instead of serializing (6 dependent HBM/L2 round trips -> 1-2).
+template <int DPL, int KT>
+__global__ void k_w2v_probe_pf(float* __restrict__ in_emb,
+                               float* __restrict__ out_emb,
+                               const long* __restrict__ in_idx,
+                               const int* __restrict__ in_off,
+                               const long* __restrict__ out_idx,
+                               const float* __restrict__ out_label,
+                               const int* __restrict__ out_off,
+                               float lr, int G, int dim) {
+  int wid = (int)((blockIdx.x * (long)blockDim.x + threadIdx.x) >> 6);
+  int lane = threadIdx.x & 63;
+  int nwaves = (int)((gridDim.x * (long)blockDim.x) >> 6);
+  for (int g = wid; g < G; g += nwaves) {
+    float h[DPL], err[DPL];
+#pragma unroll
+    for (int d = 0; d < DPL; ++d) { h[d] = 0.f; err[d] = 0.f; }
+    int ib = in_off[g], ie = in_off[g + 1];
+    for (int i = ib; i < ie; ++i) {
+      const float* row = in_emb + in_idx[i] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) h[d] += row[c];
+      }
+    }
+    int ob = out_off[g], oe = out_off[g + 1];
+    for (int t = ob; t < oe; t += KT) {
+      int kn = min(KT, oe - t);
+      float* wp[KT];
+      float wv[KT][DPL];
+      float f[KT];
+      // phase 1: issue every row load in the tile (no dependent use between)
+#pragma unroll
+      for (int k = 0; k < KT; ++k) {
+        if (k < kn) {
+          wp[k] = out_emb + out_idx[t + k] * dim;
+#pragma unroll
+          for (int d = 0; d < DPL; ++d) {
+            int c = lane + 64 * d;
+            wv[k][d] = (c < dim) ? wp[k][c] : 0.f;
+          }
+        }
+      }
+      // phase 2: dots + wave reductions
+#pragma unroll
+      for (int k = 0; k < KT; ++k) {
+        f[k] = 0.f;
+        if (k < kn) {
+#pragma unroll
+          for (int d = 0; d < DPL; ++d) f[k] += h[d] * wv[k][d];
+        }
+      }
+#pragma unroll
+      for (int s = 32; s; s >>= 1)
+#pragma unroll
+        for (int k = 0; k < KT; ++k) f[k] += __shfl_xor(f[k], s, 64);
+      // phase 3: errors + batched row writes
+#pragma unroll
+      for (int k = 0; k < KT; ++k) {
+        if (k < kn) {
+          float e = out_label[t + k] - 1.f / (1.f + expf(-f[k]));
+          float el = e * lr;
+#pragma unroll
+          for (int d = 0; d < DPL; ++d) {
+            int c = lane + 64 * d;
+            if (c < dim) {
+              err[d] += e * wv[k][d];
+              wp[k][c] = wv[k][d] + el * h[d];
+            }
+          }
+        }
+      }
+    }
+    for (int i = ib; i < ie; ++i) {
+      float* row = in_emb + in_idx[i] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) row[c] += lr * err[d];
+      }
+    }
+  }
+}
+
 int main(int argc, char** argv) {
   int G = argc > 1 ? atoi(argv[1]) : 1 << 21;
   int V = argc > 2 ? atoi(argv[2]) : 1000000;  // vocab
@@ -230,5 +317,7 @@ int main(int argc, char** argv) {
   bench(k_w2v_probe<4, 1>, "V1 plain-store  ");
   bench(k_w2v_probe_f2<2>, "V2 float2-loads ");
   bench(k_w2v_probe<4, 3>, "V3 read-only    ");
+  bench(k_w2v_probe_pf<4, 8>, "V4 prefetch-8   ");
+  bench(k_w2v_probe_pf<4, 6>, "V5 prefetch-6   ");
   return 0;
 }
