@@ -16,6 +16,10 @@ void mv_launch_sgd(float*, const float*, int64_t, hipStream_t);
 void mv_launch_momentum(float*, float*, const float*, float, int64_t, hipStream_t);
 void mv_launch_adagrad(float*, float*, const float*, float, float, float,
                        int64_t, hipStream_t);
+void mv_launch_dcasgd(float*, float*, const float*, float, float, int64_t,
+                      hipStream_t);
+void mv_launch_dcasgda(float*, float*, float*, const float*, float, float,
+                       float, float, int64_t, hipStream_t);
 void mv_launch_row_gather(float*, const float*, const int64_t*, int64_t,
                           int64_t, hipStream_t);
 void mv_launch_row_scatter_add(float*, const float*, const int64_t*, float,
@@ -79,6 +83,29 @@ void adagrad_update(torch::Tensor data, torch::Tensor gsq, torch::Tensor delta,
   mv_launch_adagrad(data.data_ptr<float>(), gsq.data_ptr<float>(),
                     delta.data_ptr<float>(), (float)lr, (float)rho, (float)eps,
                     data.numel(), cur_stream());
+}
+
+void dcasgd_update(torch::Tensor data, torch::Tensor bak, torch::Tensor delta,
+                   double lr, double lambda) {
+  check_f32(data, "data"); check_f32(bak, "bak"); check_f32(delta, "delta");
+  TORCH_CHECK(data.numel() == delta.numel() && data.numel() == bak.numel(),
+              "size mismatch");
+  mv_launch_dcasgd(data.data_ptr<float>(), bak.data_ptr<float>(),
+                   delta.data_ptr<float>(), (float)lr, (float)lambda,
+                   data.numel(), cur_stream());
+}
+
+void dcasgda_update(torch::Tensor data, torch::Tensor bak, torch::Tensor msq,
+                    torch::Tensor delta, double lr, double lambda, double rho,
+                    double eps) {
+  check_f32(data, "data"); check_f32(bak, "bak"); check_f32(msq, "msq");
+  check_f32(delta, "delta");
+  TORCH_CHECK(data.numel() == delta.numel() && data.numel() == bak.numel() &&
+              data.numel() == msq.numel(), "size mismatch");
+  mv_launch_dcasgda(data.data_ptr<float>(), bak.data_ptr<float>(),
+                    msq.data_ptr<float>(), delta.data_ptr<float>(), (float)lr,
+                    (float)lambda, (float)rho, (float)eps, data.numel(),
+                    cur_stream());
 }
 
 torch::Tensor row_gather(torch::Tensor shard, torch::Tensor rows) {
@@ -171,6 +198,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_update", &sgd_update, "K2: data -= delta");
   m.def("momentum_update", &momentum_update, "K3: fused momentum update");
   m.def("adagrad_update", &adagrad_update, "K4: fused adagrad update");
+  m.def("dcasgd_update", &dcasgd_update,
+        "DC-ASGD: delay-compensated update with per-worker backup");
+  m.def("dcasgda_update", &dcasgda_update,
+        "DC-ASGD-a: adaptive lambda via mean-square of gradients");
   m.def("row_gather", &row_gather, "K6: out[i] = shard[rows[i]]");
   m.def("row_gather_out", &row_gather_out, "K6 (preallocated out)");
   m.def("row_scatter_add", &row_scatter_add,

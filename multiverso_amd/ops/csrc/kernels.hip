@@ -136,6 +136,82 @@ __global__ void k_adagrad_tail(float* __restrict__ data, float* __restrict__ gsq
   }
 }
 
+// DC-ASGD updaters ("dcasgd"/"dcasgda", selected at reference
+// updater.cpp:51-54 from a submodule ABSENT from the snapshot
+// (.gitmodules:1-3, empty dir) — math reconstructed from the DC-ASGD
+// paper (Zheng et al., "Asynchronous SGD with Delay Compensation",
+// ICML 2017): the server compensates a delayed gradient g with a
+// diagonal Hessian approximation g*g against the backup weights the
+// worker pulled:
+//   w    -= lr * (g + lambda * g*g * (w - bak))
+//   bak   = w            (per-worker backup, updated after the step)
+// Adaptive variant keeps a mean-square m = rho*m + (1-rho)*g*g and uses
+// lambda / sqrt(m + eps) as the compensation coefficient.
+__global__ void k_dcasgd_f4(v4f* __restrict__ data, v4f* __restrict__ bak,
+                            const v4f* __restrict__ delta,
+                            float lr, float lambda, int64_t n4) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    v4f w = ntload(&data[i]), b = ntload(&bak[i]);
+    v4f g = ntload(&delta[i]);
+    w -= lr * (g + lambda * g * g * (w - b));
+    ntstore(&data[i], w);
+    ntstore(&bak[i], w);
+  }
+}
+
+__global__ void k_dcasgd_tail(float* __restrict__ data, float* __restrict__ bak,
+                              const float* __restrict__ delta,
+                              float lr, float lambda, int64_t start, int64_t n) {
+  int64_t i = start + blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    float g = delta[i];
+    float w = data[i] - lr * (g + lambda * g * g * (data[i] - bak[i]));
+    data[i] = w;
+    bak[i] = w;
+  }
+}
+
+__global__ void k_dcasgda_f4(v4f* __restrict__ data, v4f* __restrict__ bak,
+                             v4f* __restrict__ msq,
+                             const v4f* __restrict__ delta,
+                             float lr, float lambda, float rho, float eps,
+                             int64_t n4) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float om = 1.0f - rho;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    v4f w = ntload(&data[i]), b = ntload(&bak[i]);
+    v4f g = ntload(&delta[i]);
+    v4f m = rho * ntload(&msq[i]) + om * g * g;
+    ntstore(&msq[i], m);
+    v4f lam;
+    lam.x = lambda * __frsqrt_rn(m.x + eps);
+    lam.y = lambda * __frsqrt_rn(m.y + eps);
+    lam.z = lambda * __frsqrt_rn(m.z + eps);
+    lam.w = lambda * __frsqrt_rn(m.w + eps);
+    w -= lr * (g + lam * g * g * (w - b));
+    ntstore(&data[i], w);
+    ntstore(&bak[i], w);
+  }
+}
+
+__global__ void k_dcasgda_tail(float* __restrict__ data, float* __restrict__ bak,
+                               float* __restrict__ msq,
+                               const float* __restrict__ delta,
+                               float lr, float lambda, float rho, float eps,
+                               int64_t start, int64_t n) {
+  int64_t i = start + blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    float g = delta[i];
+    float m = rho * msq[i] + (1.0f - rho) * g * g;
+    msq[i] = m;
+    float w = data[i]
+        - lr * (g + lambda * __frsqrt_rn(m + eps) * g * g * (data[i] - bak[i]));
+    data[i] = w;
+    bak[i] = w;
+  }
+}
+
 __global__ void k_copy_f4(v4f* __restrict__ dst,
                           const v4f* __restrict__ src, int64_t n4) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -245,6 +321,28 @@ void mv_launch_adagrad(float* data, float* gsq, const float* delta,
   int64_t tail = n - n4 * 4;
   if (tail) k_adagrad_tail<<<1, 64, 0, s>>>(data, gsq, delta, inv_lr, rho, eps,
                                             n4 * 4, n);
+}
+
+void mv_launch_dcasgd(float* data, float* bak, const float* delta,
+                      float lr, float lambda, int64_t n, hipStream_t s) {
+  int64_t n4 = n / 4;
+  if (n4) k_dcasgd_f4<<<grid_for_cap(n4, ELEM_GRID), BLOCK, 0, s>>>(
+      (v4f*)data, (v4f*)bak, (const v4f*)delta, lr, lambda, n4);
+  int64_t tail = n - n4 * 4;
+  if (tail) k_dcasgd_tail<<<1, 64, 0, s>>>(data, bak, delta, lr, lambda,
+                                           n4 * 4, n);
+}
+
+void mv_launch_dcasgda(float* data, float* bak, float* msq, const float* delta,
+                       float lr, float lambda, float rho, float eps,
+                       int64_t n, hipStream_t s) {
+  int64_t n4 = n / 4;
+  if (n4) k_dcasgda_f4<<<grid_for_cap(n4, ELEM_GRID), BLOCK, 0, s>>>(
+      (v4f*)data, (v4f*)bak, (v4f*)msq, (const v4f*)delta, lr, lambda, rho,
+      eps, n4);
+  int64_t tail = n - n4 * 4;
+  if (tail) k_dcasgda_tail<<<1, 64, 0, s>>>(data, bak, msq, delta, lr, lambda,
+                                            rho, eps, n4 * 4, n);
 }
 
 void mv_launch_row_gather(float* out, const float* shard, const int64_t* rows,

@@ -132,11 +132,80 @@ class AdaGradUpdater(Updater):
             self.shard.sub_(rho * g / torch.sqrt(self.g_sqr + self.EPS))
 
 
+class DCASGDUpdater(Updater):
+    """DC-ASGD ("dcasgd", reference updater.cpp:51 — the updater itself
+    lives in a submodule absent from the snapshot; math from Zheng et al.,
+    "Asynchronous SGD with Delay Compensation", ICML 2017):
+
+        w   -= lr * (g + lambda * g*g * (w - bak_k))
+        bak_k = w
+
+    ``bak_k`` is the per-worker backup of the weights worker k last
+    synchronized (AddOption.worker_id selects the slot; worker_id < 0 —
+    the collective merged-delta lane — uses one shared slot). Backups are
+    lazily initialized to the shard's current values."""
+
+    name = "dcasgd"
+
+    def __init__(self, shard: torch.Tensor) -> None:
+        super().__init__(shard)
+        self._bak: Dict[int, torch.Tensor] = {}
+
+    def _backup(self, worker_id: int) -> torch.Tensor:
+        b = self._bak.get(worker_id)
+        if b is None:
+            b = self.shard.detach().clone()
+            self._bak[worker_id] = b
+        return b
+
+    def update(self, delta: torch.Tensor, option: Optional[AddOption]) -> None:
+        opt = option or AddOption()
+        lr, lam = opt.learning_rate, opt.lambda_
+        bak = self._backup(opt.worker_id)
+        if self.shard.is_cuda:
+            _hip_ops().dcasgd_update(self.shard, bak, delta,
+                                     float(lr), float(lam))
+        else:
+            g = delta
+            self.shard.sub_(lr * (g + lam * g * g * (self.shard - bak)))
+            bak.copy_(self.shard)
+
+
+class DCASGDAUpdater(DCASGDUpdater):
+    """DC-ASGD-a ("dcasgda", updater.cpp:52): adaptive lambda — a running
+    mean-square of the gradient m = rho*m + (1-rho)*g*g scales the
+    compensation coefficient to lambda / sqrt(m + eps)."""
+
+    name = "dcasgda"
+    EPS = 1e-7
+
+    def __init__(self, shard: torch.Tensor) -> None:
+        super().__init__(shard)
+        self.mean_sqr = torch.zeros_like(shard)
+
+    def update(self, delta: torch.Tensor, option: Optional[AddOption]) -> None:
+        opt = option or AddOption()
+        lr, lam, rho = opt.learning_rate, opt.lambda_, opt.rho
+        bak = self._backup(opt.worker_id)
+        if self.shard.is_cuda:
+            _hip_ops().dcasgda_update(self.shard, bak, self.mean_sqr, delta,
+                                      float(lr), float(lam), float(rho),
+                                      self.EPS)
+        else:
+            g = delta
+            self.mean_sqr.mul_(rho).add_((1.0 - rho) * g * g)
+            lam_t = lam / torch.sqrt(self.mean_sqr + self.EPS)
+            self.shard.sub_(lr * (g + lam_t * g * g * (self.shard - bak)))
+            bak.copy_(self.shard)
+
+
 _REGISTRY: Dict[str, Type[Updater]] = {
     "default": Updater,
     "sgd": SGDUpdater,
     "momentum": MomentumUpdater,
     "adagrad": AdaGradUpdater,
+    "dcasgd": DCASGDUpdater,
+    "dcasgda": DCASGDAUpdater,
 }
 
 

@@ -63,6 +63,32 @@ def test_adagrad(hip, n):
     assert torch.allclose(data, ref_d, rtol=1e-4, atol=1e-5)
 
 
+@pytest.mark.parametrize("n", [64, 999, 1 << 18])
+def test_dcasgd(hip, n):
+    data, bak, delta = rand(n, 20), rand(n, 21), rand(n, 22)
+    lr, lam = 0.1, 0.04
+    ref = data - lr * (delta + lam * delta * delta * (data - bak))
+    hip.dcasgd_update(data, bak, delta, lr, lam)
+    torch.cuda.synchronize()
+    assert torch.allclose(data, ref, rtol=1e-5, atol=1e-6)
+    assert torch.equal(bak, data)
+
+
+@pytest.mark.parametrize("n", [64, 999, 1 << 18])
+def test_dcasgda(hip, n):
+    data, bak, msq, delta = rand(n, 23), rand(n, 24), rand(n, 25).abs(), \
+        rand(n, 26)
+    lr, lam, rho, eps = 0.1, 0.04, 0.95, 1e-7
+    ref_m = rho * msq + (1 - rho) * delta * delta
+    lam_t = lam / torch.sqrt(ref_m + eps)
+    ref = data - lr * (delta + lam_t * delta * delta * (data - bak))
+    hip.dcasgda_update(data, bak, msq, delta, lr, lam, rho, eps)
+    torch.cuda.synchronize()
+    assert torch.allclose(msq, ref_m, rtol=1e-5, atol=1e-7)
+    assert torch.allclose(data, ref, rtol=1e-4, atol=1e-5)
+    assert torch.equal(bak, data)
+
+
 @pytest.mark.parametrize("cols", [128, 200, 7])
 def test_row_gather(hip, cols):
     shard = rand(500 * cols, 11).view(500, cols)

@@ -109,6 +109,37 @@ def test_adagrad_updater(env):
     assert torch.allclose(t.get(), torch.full((4,), float(expect)))
 
 
+def test_dcasgd_updater(env):
+    t = mv.ArrayTable(4, updater_type="dcasgd")
+    opt = mv.AddOption(learning_rate=0.1, lambda_=0.5)
+    g = torch.full((4,), 2.0)
+    # backup lazily = current shard (0); first step: w -= 0.1*(2 + 0.5*4*0)
+    t.add(g, option=opt)
+    assert torch.allclose(t.get(), torch.full((4,), -0.2))
+    # second step from a different worker: its backup initializes to the
+    # CURRENT w (-0.2), so the compensation term is again 0
+    opt2 = mv.AddOption(worker_id=1, learning_rate=0.1, lambda_=0.5)
+    t.add(g, option=opt2)
+    assert torch.allclose(t.get(), torch.full((4,), -0.4))
+    # third step from worker 0: bak_0 = -0.2, w = -0.4
+    # w -= 0.1 * (2 + 0.5*4*(-0.4 - -0.2)) = -0.4 - 0.1*(2 - 0.4) = -0.56
+    t.add(g, option=opt)
+    assert torch.allclose(t.get(), torch.full((4,), -0.56))
+
+
+def test_dcasgda_updater(env):
+    t = mv.ArrayTable(4, updater_type="dcasgda")
+    opt = mv.AddOption(learning_rate=0.1, lambda_=0.5, rho=0.9)
+    g = torch.full((4,), 2.0)
+    t.add(g, option=opt)
+    # m = 0.1*4 = 0.4; lam_t = 0.5/sqrt(0.4+eps); bak=w0=0 -> comp term 0
+    assert torch.allclose(t.get(), torch.full((4,), -0.2))
+    t.add(g, option=opt)
+    # m = 0.9*0.4 + 0.1*4 = 0.76; lam_t = 0.5/sqrt(0.76); bak=-0.2, w=-0.2
+    # comp = lam_t*4*(w - bak) = 0 -> w = -0.4
+    assert torch.allclose(t.get(), torch.full((4,), -0.4))
+
+
 def test_checkpoint_roundtrip(env, tmp_path):
     t = mv.ArrayTable(32)
     t.add(torch.arange(32, dtype=torch.float32))
