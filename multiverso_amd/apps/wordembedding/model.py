@@ -93,6 +93,7 @@ class WordEmbedding:
             self._build_hs_tensors()
         self.learning_rate = option.init_learning_rate
         self._block_pool: Optional[torch.Tensor] = None
+        self._seed_state = (option.seed * 0x2545F4914F6CDD1D) & 0xFFFFFFFFFFFFFFFF
         self.word_count_local = 0      # words since last global sync
         self.word_count_actual = 0     # global processed words
         self.gen = torch.Generator(device=self.device)
@@ -114,22 +115,21 @@ class WordEmbedding:
     # ------------------------------------------------------------------
     # Group construction (PrepareData equivalent, fully batched on GPU)
     # ------------------------------------------------------------------
-    def build_groups(self, words: torch.Tensor, sent_ids: torch.Tensor):
-        """Returns (in_idx, in_off, out_idx, out_label, out_off) with
-        GLOBAL word/node ids; groups follow word2vec semantics:
-        skip-gram: input = context word, outputs = center (+negs/path);
-        CBOW: inputs = context words of a center, outputs = center (+...).
-        """
+    def build_pairs(self, words: torch.Tensor, sent_ids: torch.Tensor):
+        """Returns (in_idx, in_off, centers) with GLOBAL word ids: one
+        group per training sample, inputs per group via in_off ragged
+        offsets, `centers` the positive output of each group."""
         opt = self.opt
         if opt.cbow:
-            return self._build_cbow_groups(words, sent_ids)
+            return self._build_cbow_pairs(words, sent_ids)
         device = self.device
         n = words.numel()
         W = min(opt.window, n - 1)
+        empty = (torch.empty(0, dtype=torch.int64, device=device),
+                 torch.zeros(1, dtype=torch.int32, device=device),
+                 torch.empty(0, dtype=torch.int64, device=device))
         if W < 1:
-            e = torch.empty(0, dtype=torch.int64, device=device)
-            z = torch.zeros(1, dtype=torch.int32, device=device)
-            return e, z, e, e.float(), z
+            return empty
         # all window offsets in one batched pass: pair (i, i+o) for
         # o = 1..W, valid when in-range + same sentence + reduced-window
         # keep (offset o survives with prob (window-o+1)/window)
@@ -145,31 +145,41 @@ class WordEmbedding:
         li = left.expand(n, W).reshape(-1)[m]
         ri = right.reshape(-1)[m]
         if li.numel() == 0:
-            e = torch.empty(0, dtype=torch.int64, device=device)
-            z = torch.zeros(1, dtype=torch.int32, device=device)
-            return e, z, e, e.float(), z
+            return empty
         # center at i, context at i+o  AND  center at i+o, context at i
         wl, wr = words[li], words[ri]
         centers = torch.cat([wl, wr])
         contexts = torch.cat([wr, wl])
-
         # skip-gram: one group per pair, input = context word
         g = centers.numel()
-        in_idx = contexts
         in_off = torch.arange(g + 1, dtype=torch.int32, device=device)
+        return contexts, in_off, centers
+
+    def build_groups(self, words: torch.Tensor, sent_ids: torch.Tensor):
+        """Returns (in_idx, in_off, out_idx, out_label, out_off) with
+        GLOBAL word/node ids; groups follow word2vec semantics:
+        skip-gram: input = context word, outputs = center (+negs/path);
+        CBOW: inputs = context words of a center, outputs = center (+...).
+        """
+        in_idx, in_off, centers = self.build_pairs(words, sent_ids)
+        if centers.numel() == 0:
+            e = torch.empty(0, dtype=torch.int64, device=self.device)
+            z = torch.zeros(1, dtype=torch.int32, device=self.device)
+            return e, z, e, e.float(), z
         out_idx, out_label, out_off = self._outputs_for(centers)
         return in_idx, in_off, out_idx, out_label, out_off
 
-    def _build_cbow_groups(self, words: torch.Tensor,
+    def _build_cbow_pairs(self, words: torch.Tensor,
                            sent_ids: torch.Tensor):
         opt = self.opt
         device = self.device
         n = words.numel()
         W = min(opt.window, n - 1)
+        empty = (torch.empty(0, dtype=torch.int64, device=device),
+                 torch.zeros(1, dtype=torch.int32, device=device),
+                 torch.empty(0, dtype=torch.int64, device=device))
         if W < 1:
-            e = torch.empty(0, dtype=torch.int64, device=device)
-            z = torch.zeros(1, dtype=torch.int32, device=device)
-            return e, z, e, e.float(), z
+            return empty
         offs = torch.arange(1, W + 1, device=device)
         left = torch.arange(n, device=device).unsqueeze(1)
         right = left + offs
@@ -182,21 +192,16 @@ class WordEmbedding:
         li = left.expand(n, W).reshape(-1)[m]
         ri = right.reshape(-1)[m]
         if li.numel() == 0:
-            e = torch.empty(0, dtype=torch.int64, device=device)
-            z = torch.zeros(1, dtype=torch.int32, device=device)
-            return e, z, e, e.float(), z
+            return empty
         pos = torch.cat([li, ri])      # center positions
         ctx = torch.cat([words[ri], words[li]])
         order = torch.argsort(pos, stable=True)
         pos, ctx = pos[order], ctx[order]
         upos, counts = torch.unique_consecutive(pos, return_counts=True)
-        in_idx = ctx
         in_off = torch.zeros(upos.numel() + 1, dtype=torch.int32,
                              device=device)
         in_off[1:] = counts.cumsum(0).to(torch.int32)
-        centers = words[upos]
-        out_idx, out_label, out_off = self._outputs_for(centers)
-        return in_idx, in_off, out_idx, out_label, out_off
+        return ctx, in_off, words[upos]
 
     def _outputs_for(self, centers: torch.Tensor):
         """Output node/label lists per group for NS and/or HS. Negatives
@@ -279,10 +284,23 @@ class WordEmbedding:
             # (wordembedding.cpp:208)
             self._block_pool = torch.unique(self.sampler.negative_sampling(
                 (opt.negative_num * uwords.numel(),), generator=self.gen))
-        in_idx, in_off, out_idx, out_label, out_off = \
-            self.build_groups(words, sent_ids)
-        if in_idx.numel() == 0 or out_idx.numel() == 0:
-            return int(words.numel())
+
+        # NS fast path (GPU, negative sampling without HS): negatives are
+        # generated INSIDE the kernel from the block pool with the
+        # reference's LCG scheme (wordembedding.cpp:276-279) — no
+        # host-side randint/gather/label/ragged build at all.
+        ns_fast = (opt.negative_num > 0 and not opt.hs
+                   and self.device.type == "cuda")
+        if ns_fast:
+            in_idx, in_off, centers = self.build_pairs(words, sent_ids)
+            if centers.numel() == 0:
+                return int(words.numel())
+            out_idx = out_label = out_off = None
+        else:
+            in_idx, in_off, out_idx, out_label, out_off = \
+                self.build_groups(words, sent_ids)
+            if in_idx.numel() == 0 or out_idx.numel() == 0:
+                return int(words.numel())
 
         if mv.size() == 1:
             # Single-rank fast path: every row is local, and
@@ -293,10 +311,16 @@ class WordEmbedding:
             if opt.use_adagrad:
                 igq = self.input_gsq_table.shard
                 ogq = self.output_gsq_table.shard
-            self._train_kernel(self.input_table.shard,
-                               self.output_table.shard, igq, ogq,
-                               in_idx, in_off, out_idx,
-                               out_label.float(), out_off)
+            if ns_fast:
+                self._train_kernel_ns(self.input_table.shard,
+                                      self.output_table.shard, igq, ogq,
+                                      in_idx, in_off, centers,
+                                      self._block_pool)
+            else:
+                self._train_kernel(self.input_table.shard,
+                                   self.output_table.shard, igq, ogq,
+                                   in_idx, in_off, out_idx,
+                                   out_label.float(), out_off)
             nwords = int(words.numel())
             self._update_lr(nwords)
             return nwords
@@ -315,7 +339,11 @@ class WordEmbedding:
             cand.append(self._hs_point[
                 starts.repeat_interleave(lens) + _segment_arange(lens)])
         uout = torch.unique(torch.cat(cand))
-        out_local = torch.searchsorted(uout, out_idx)
+        if ns_fast:
+            out_local = torch.searchsorted(uout, centers)
+            pool_local = torch.searchsorted(uout, self._block_pool)
+        else:
+            out_local = torch.searchsorted(uout, out_idx)
 
         # pull touched rows (RequestParameter)
         in_buf = self.input_table.get_rows(uin).contiguous()
@@ -328,9 +356,13 @@ class WordEmbedding:
             ogq = self.output_gsq_table.get_rows(uout).contiguous()
             gbufs = (igq, igq.clone(), ogq, ogq.clone())
 
-        self._train_kernel(in_buf, out_buf, gbufs[0], gbufs[2],
-                           in_local, in_off, out_local,
-                           out_label.float(), out_off)
+        if ns_fast:
+            self._train_kernel_ns(in_buf, out_buf, gbufs[0], gbufs[2],
+                                  in_local, in_off, out_local, pool_local)
+        else:
+            self._train_kernel(in_buf, out_buf, gbufs[0], gbufs[2],
+                               in_local, in_off, out_local,
+                               out_label.float(), out_off)
 
         # push deltas (AddDeltaParameter, /num_workers)
         p = float(mv.workers_num())
@@ -343,6 +375,46 @@ class WordEmbedding:
         nwords = int(words.numel())
         self._update_lr(nwords)
         return nwords
+
+    def _next_seed(self) -> int:
+        """Per-launch seed for the in-kernel negative LCG: a host-side
+        splitmix step — no device sync, reproducible from opt.seed."""
+        s = (self._seed_state + 0x9E3779B97F4A7C15) & 0xFFFFFFFFFFFFFFFF
+        self._seed_state = s
+        z = s
+        z = ((z ^ (z >> 30)) * 0xBF58476D1CE4E5B9) & 0xFFFFFFFFFFFFFFFF
+        z = ((z ^ (z >> 27)) * 0x94D049BB133111EB) & 0xFFFFFFFFFFFFFFFF
+        return (z ^ (z >> 31)) & 0x7FFFFFFFFFFFFFFF
+
+    def _train_kernel_ns(self, in_buf, out_buf, in_gsq, out_gsq,
+                         in_local, in_off, centers, pool):
+        """NS fast path: negatives generated in-kernel from `pool`."""
+        from ... import ops
+        hip = ops.module(required=True)
+        dummy = in_buf
+        igq = in_gsq if in_gsq is not None else dummy
+        ogq = out_gsq if out_gsq is not None else dummy
+        in_off = in_off.to(torch.int32)
+        g_total = in_off.numel() - 1
+        step = self.opt.max_groups_per_launch
+        if g_total <= step:
+            hip.w2v_train_ns(in_buf, out_buf, igq, ogq, in_local, in_off,
+                             centers, pool, self.opt.negative_num,
+                             self._next_seed(), self.learning_rate,
+                             self.opt.use_adagrad, self.opt.atomic_updates)
+            return
+        gs = list(range(0, g_total, step)) + [g_total]
+        ibs = in_off[gs].tolist()
+        for k in range(len(gs) - 1):
+            g0, g1 = gs[k], gs[k + 1]
+            i0, i1 = ibs[k], ibs[k + 1]
+            hip.w2v_train_ns(in_buf, out_buf, igq, ogq,
+                             in_local[i0:i1].contiguous(),
+                             (in_off[g0:g1 + 1] - i0).contiguous(),
+                             centers[g0:g1].contiguous(), pool,
+                             self.opt.negative_num, self._next_seed(),
+                             self.learning_rate, self.opt.use_adagrad,
+                             self.opt.atomic_updates)
 
     def _train_kernel(self, in_buf, out_buf, in_gsq, out_gsq,
                       in_local, in_off, out_local, out_label, out_off):

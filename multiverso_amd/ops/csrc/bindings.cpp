@@ -27,6 +27,10 @@ void mv_launch_row_scatter_add(float*, const float*, const int64_t*, float,
 void mv_launch_w2v(float*, float*, float*, float*, const int64_t*, const int*,
                    const int64_t*, const float*, const int*, float, int64_t,
                    int64_t, int, int, hipStream_t);
+void mv_launch_w2v_ns(float*, float*, float*, float*, const int64_t*,
+                      const int*, const int64_t*, const int64_t*, int64_t,
+                      int, uint64_t, float, int64_t, int64_t, int, int,
+                      hipStream_t);
 void mv_launch_row_scatter_adagrad(float*, float*, const float*,
                                    const int64_t*, float, float, float,
                                    int64_t, int64_t, hipStream_t);
@@ -190,6 +194,39 @@ void w2v_train(torch::Tensor in_emb, torch::Tensor out_emb,
                 use_adagrad ? 1 : 0, use_atomic ? 1 : 0, cur_stream());
 }
 
+void w2v_train_ns(torch::Tensor in_emb, torch::Tensor out_emb,
+                  torch::Tensor in_gsq, torch::Tensor out_gsq,
+                  torch::Tensor in_idx, torch::Tensor in_off,
+                  torch::Tensor centers, torch::Tensor pool, int64_t neg,
+                  int64_t seed, double lr, bool use_adagrad,
+                  bool use_atomic) {
+  check_f32(in_emb, "in_emb"); check_f32(out_emb, "out_emb");
+  TORCH_CHECK(in_emb.dim() == 2 && out_emb.dim() == 2, "emb must be 2-D");
+  TORCH_CHECK(in_emb.size(1) == out_emb.size(1), "dim mismatch");
+  TORCH_CHECK(in_emb.size(1) <= 512, "w2v kernel supports dim <= 512");
+  TORCH_CHECK(in_idx.scalar_type() == torch::kInt64 &&
+              centers.scalar_type() == torch::kInt64 &&
+              pool.scalar_type() == torch::kInt64, "ids must be int64");
+  TORCH_CHECK(in_off.scalar_type() == torch::kInt32, "offsets must be int32");
+  TORCH_CHECK(pool.numel() > 0, "empty negative pool");
+  int64_t G = in_off.numel() - 1;
+  TORCH_CHECK(centers.numel() == G, "centers/group count mismatch");
+  float *igq = nullptr, *ogq = nullptr;
+  if (use_adagrad) {
+    check_f32(in_gsq, "in_gsq"); check_f32(out_gsq, "out_gsq");
+    TORCH_CHECK(in_gsq.sizes() == in_emb.sizes() &&
+                out_gsq.sizes() == out_emb.sizes(), "gsq shape mismatch");
+    igq = in_gsq.data_ptr<float>();
+    ogq = out_gsq.data_ptr<float>();
+  }
+  mv_launch_w2v_ns(in_emb.data_ptr<float>(), out_emb.data_ptr<float>(), igq,
+                   ogq, in_idx.data_ptr<int64_t>(), in_off.data_ptr<int>(),
+                   centers.data_ptr<int64_t>(), pool.data_ptr<int64_t>(),
+                   pool.numel(), (int)neg, (uint64_t)seed, (float)lr, G,
+                   in_emb.size(1), use_adagrad ? 1 : 0, use_atomic ? 1 : 0,
+                   cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -211,4 +248,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("w2v_train", &w2v_train,
         "K9-K11: fused word2vec block training (skip-gram/CBOW, NS/HS, "
         "optional adagrad)");
+  m.def("w2v_train_ns", &w2v_train_ns,
+        "K9-K11 NS fast path: negatives generated in-kernel from the "
+        "per-block pool (reference LCG scheme)");
 }

@@ -479,6 +479,116 @@ __global__ void k_w2v(float* __restrict__ in_emb, float* __restrict__ out_emb,
   }
 }
 
+// Negative-sampling variant with IN-KERNEL negative generation: for
+// NS-only training (no hierarchical softmax) the output list of group g
+// is implicit — 1 positive (centers[g], label 1) + `neg` draws from the
+// per-block pool with the reference's own LCG scheme
+// (wordembedding.cpp:276-279: next_random = next_random*25214903917+11;
+// index = (next_random >> 8) % pool_size; skip if == positive). This
+// removes the host-side randint/gather/label/ragged-offset build and
+// shrinks the per-group index traffic from (1+neg) ids to 1.
+template <int DPL, bool ADAGRAD, bool ATOMIC>
+__global__ void k_w2v_ns(float* __restrict__ in_emb, float* __restrict__ out_emb,
+                         float* __restrict__ in_gsq, float* __restrict__ out_gsq,
+                         const int64_t* __restrict__ in_idx,
+                         const int* __restrict__ in_off,
+                         const int64_t* __restrict__ centers,
+                         const int64_t* __restrict__ pool, int64_t pool_n,
+                         int neg, uint64_t seed,
+                         float lr, int G, int dim) {
+  int wid = (int)((blockIdx.x * (int64_t)blockDim.x + threadIdx.x) >> 6);
+  int lane = threadIdx.x & 63;
+  int nwaves = (int)((gridDim.x * (int64_t)blockDim.x) >> 6);
+  for (int g = wid; g < G; g += nwaves) {
+    float h[DPL], err[DPL];
+#pragma unroll
+    for (int d = 0; d < DPL; ++d) { h[d] = 0.f; err[d] = 0.f; }
+    int ib = in_off[g], ie = in_off[g + 1];
+    for (int i = ib; i < ie; ++i) {
+      const float* row = in_emb + in_idx[i] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) h[d] += row[c];
+      }
+    }
+    if (ie - ib > 1) {
+      float inv = 1.f / (float)(ie - ib);
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) h[d] *= inv;
+    }
+    int64_t pos = centers[g];
+    // per-group random stream (same recurrence as util.cpp:144)
+    uint64_t next_random = seed + (uint64_t)g * 25214903917ull + 11ull;
+    for (int o = 0; o <= neg; ++o) {
+      int64_t node;
+      float label;
+      if (o == 0) {
+        node = pos; label = 1.f;
+      } else {
+        next_random = next_random * 25214903917ull + 11ull;
+        node = pool[(int64_t)((next_random >> 8) % (uint64_t)pool_n)];
+        label = 0.f;
+        if (node == pos) continue;  // wordembedding.cpp:279
+      }
+      float* w = out_emb + node * dim;
+      float wv[DPL];
+      float f = 0.f;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        wv[d] = (c < dim) ? w[c] : 0.f;
+        f += h[d] * wv[d];
+      }
+#pragma unroll
+      for (int s = 32; s; s >>= 1) f += __shfl_xor(f, s, 64);
+      f = 1.f / (1.f + expf(-f));
+      float e = label - f;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) {
+          err[d] += e * wv[d];
+          if (ADAGRAD) {
+            float gg = e * h[d];
+            float* gq = out_gsq + node * dim + c;
+            float G2;
+            if (ATOMIC) G2 = atomicAdd(gq, gg * gg) + gg * gg;
+            else { G2 = *gq + gg * gg; *gq = G2; }
+            float dw = (G2 > 1e-10f) ? gg * lr * __frsqrt_rn(G2) : 0.f;
+            if (ATOMIC) atomicAdd(&w[c], dw);
+            else w[c] = wv[d] + dw;
+          } else {
+            if (ATOMIC) atomicAdd(&w[c], e * lr * h[d]);
+            else w[c] = wv[d] + e * lr * h[d];
+          }
+        }
+      }
+    }
+    for (int i = ib; i < ie; ++i) {
+      float* row = in_emb + in_idx[i] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) {
+          if (ADAGRAD) {
+            float* gq = in_gsq + in_idx[i] * dim + c;
+            float G2;
+            if (ATOMIC) G2 = atomicAdd(gq, err[d] * err[d]) + err[d] * err[d];
+            else { G2 = *gq + err[d] * err[d]; *gq = G2; }
+            float dw = (G2 > 1e-10f) ? err[d] * lr * __frsqrt_rn(G2) : 0.f;
+            if (ATOMIC) atomicAdd(&row[c], dw);
+            else row[c] += dw;
+          } else {
+            if (ATOMIC) atomicAdd(&row[c], lr * err[d]);
+            else row[c] += lr * err[d];
+          }
+        }
+      }
+    }
+  }
+}
+
 // Hogwild (plain-store) row updates by default — the reference's own
 // unsynchronized OpenMP trainers race identically (wordembedding.cpp
 // trainers share rows with no locks); on MI355X plain stores measured
@@ -515,6 +625,40 @@ extern "C" void mv_launch_w2v(float* in_emb, float* out_emb,
   }
 #undef W2V_CASE
 #undef W2V_LAUNCH
+}
+
+extern "C" void mv_launch_w2v_ns(float* in_emb, float* out_emb,
+                                 float* in_gsq, float* out_gsq,
+                                 const int64_t* in_idx, const int* in_off,
+                                 const int64_t* centers,
+                                 const int64_t* pool, int64_t pool_n,
+                                 int neg, uint64_t seed, float lr, int64_t G,
+                                 int64_t dim, int use_adagrad, int use_atomic,
+                                 hipStream_t s) {
+  if (!G) return;
+  int grid = grid_for(G * 64);
+  int dpl = (int)((dim + 63) / 64);
+#define W2VNS_LAUNCH(D, A, AT)                                               \
+  k_w2v_ns<D, A, AT><<<grid, BLOCK, 0, s>>>(in_emb, out_emb, in_gsq,         \
+      out_gsq, in_idx, in_off, centers, pool, pool_n, neg, seed, lr,         \
+      (int)G, (int)dim)
+#define W2VNS_CASE(D)                                                        \
+  case D:                                                                    \
+    if (use_adagrad) {                                                       \
+      if (use_atomic) W2VNS_LAUNCH(D, true, true);                           \
+      else W2VNS_LAUNCH(D, true, false);                                     \
+    } else {                                                                 \
+      if (use_atomic) W2VNS_LAUNCH(D, false, true);                          \
+      else W2VNS_LAUNCH(D, false, false);                                    \
+    }                                                                        \
+    break;
+  switch (dpl) {
+    W2VNS_CASE(1) W2VNS_CASE(2) W2VNS_CASE(3) W2VNS_CASE(4)
+    W2VNS_CASE(5) W2VNS_CASE(6) W2VNS_CASE(7) W2VNS_CASE(8)
+    default: break;  // dim > 512 unsupported by this kernel
+  }
+#undef W2VNS_CASE
+#undef W2VNS_LAUNCH
 }
 
 // ---------------------------------------------------------------------------

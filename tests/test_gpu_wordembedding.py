@@ -71,6 +71,66 @@ def test_w2v_kernel_adagrad_vs_reference():
             (got.cpu() - want).abs().max()
 
 
+def test_w2v_ns_kernel_deterministic_pool_vs_reference():
+    """NS fast path with a 1-element pool and a single group: every
+    in-kernel negative is the known pool row, so the output sequence is
+    exactly (positive, neg x pool-row) and — with one wave and plain
+    stores (own-thread coherent) — the kernel must match the sequential
+    torch reference to fp32 tolerance."""
+    from multiverso_amd import ops
+    from multiverso_amd.apps.wordembedding.model import _w2v_train_torch
+    hip = ops.module(required=True)
+    dim, neg = 128, 3
+    torch.manual_seed(4)
+    in_buf = torch.randn(1, dim) * 0.1
+    out_buf = torch.randn(2, dim) * 0.1
+    in_idx = torch.zeros(1, dtype=torch.int64)
+    in_off = torch.tensor([0, 1], dtype=torch.int32)
+    centers = torch.zeros(1, dtype=torch.int64)     # positive = out row 0
+    pool = torch.tensor([1], dtype=torch.int64)     # negatives = out row 1
+    lr = 0.05
+
+    ref_in, ref_out = in_buf.clone(), out_buf.clone()
+    out_idx = torch.tensor([0] + [1] * neg, dtype=torch.int64)
+    lab = torch.tensor([1.0] + [0.0] * neg)
+    out_off = torch.tensor([0, 1 + neg], dtype=torch.int32)
+    _w2v_train_torch(ref_in, ref_out, None, None, in_idx, in_off,
+                     out_idx, lab, out_off, lr, False, lr)
+
+    d = lambda t: t.cuda()
+    gin, gout = d(in_buf), d(out_buf)
+    hip.w2v_train_ns(gin, gout, gin, gout, d(in_idx), d(in_off), d(centers),
+                     d(pool), neg, 12345, lr, False, False)
+    torch.cuda.synchronize()
+    assert torch.allclose(gin.cpu(), ref_in, rtol=1e-4, atol=1e-5), \
+        (gin.cpu() - ref_in).abs().max()
+    assert torch.allclose(gout.cpu(), ref_out, rtol=1e-4, atol=1e-5), \
+        (gout.cpu() - ref_out).abs().max()
+
+
+def test_w2v_ns_kernel_only_touches_pool_and_centers():
+    """In-kernel negatives must come from the pool: rows outside
+    centers ∪ pool stay bit-identical."""
+    from multiverso_amd import ops
+    hip = ops.module(required=True)
+    g, dim, V = 8, 64, 64
+    torch.manual_seed(5)
+    in_buf = torch.randn(V, dim).cuda() * 0.1
+    out_buf = torch.randn(V, dim).cuda() * 0.1
+    before = out_buf.clone()
+    in_idx = torch.arange(g, dtype=torch.int64).cuda()
+    in_off = torch.arange(g + 1, dtype=torch.int32).cuda()
+    centers = torch.arange(10, 10 + g, dtype=torch.int64).cuda()
+    pool = torch.tensor([40, 41, 42], dtype=torch.int64).cuda()
+    hip.w2v_train_ns(out_buf, out_buf, out_buf, out_buf, in_idx, in_off,
+                     centers, pool, 5, 999, 0.1, False, False)
+    torch.cuda.synchronize()
+    touched = set(range(g)) | set(range(10, 10 + g)) | {40, 41, 42}
+    untouched = [r for r in range(V) if r not in touched]
+    assert torch.equal(out_buf[untouched], before[untouched])
+    assert not torch.equal(out_buf[list(touched)], before[list(touched)])
+
+
 def test_w2v_gpu_end_to_end_learns():
     import multiverso_amd as mv
     from multiverso_amd.apps.wordembedding.model import (WordEmbedding,
