@@ -24,6 +24,7 @@ import numpy as np
 import torch
 
 import multiverso_amd as mv
+from multiverso_amd import ops
 
 from .objective import Batch, create_objective
 
@@ -59,10 +60,10 @@ class LocalModel:
         grad, loss = self.objective.gradient(batch, w_rows)
         if self.cfg.objective_type == "ftrl":
             # server/local ftrl update is unscaled: state -= delta
-            self.weight.index_add_(0, batch.keys, -grad)
+            ops.scatter_add_rows(self.weight, batch.keys, grad, -1.0)
         else:
             lr = self.sched.next_lr()
-            self.weight.index_add_(0, batch.keys, -lr * grad)
+            ops.scatter_add_rows(self.weight, batch.keys, grad, -lr)
         return loss
 
     def predict(self, batch: Batch) -> torch.Tensor:
@@ -131,15 +132,15 @@ class PSModel:
             grad, loss = self.objective.gradient(b, w_rows)
             total_loss += loss
             if self.is_ftrl:
-                local.index_add_(0, lidx, -grad)
+                ops.scatter_add_rows(local, lidx, grad, -1.0)
             elif adagrad:
                 # server-side adagrad consumes lr-scaled deltas; locally
                 # approximate with plain sgd steps for within-chunk vis.
                 lr = self.sched.next_lr()
-                local.index_add_(0, lidx, -lr * grad)
+                ops.scatter_add_rows(local, lidx, grad, -lr)
             else:
                 lr = self.sched.next_lr()
-                local.index_add_(0, lidx, -lr * grad)
+                ops.scatter_add_rows(local, lidx, grad, -lr)
         if adagrad:
             delta = pulled - local   # = sum(lr*grad); server g=delta/lr
             opt = mv.AddOption(learning_rate=1.0, rho=self.cfg.learning_rate)

@@ -18,6 +18,8 @@ from typing import Optional, Tuple
 
 import torch
 
+from multiverso_amd import ops
+
 
 class Batch:
     """Sparse minibatch: sample i has keys[ptr[i]:ptr[i+1]] etc."""
@@ -54,7 +56,7 @@ def _scores(batch: Batch, w_rows: torch.Tensor) -> torch.Tensor:
     B, O = batch.size, w_rows.size(1)
     contrib = batch.vals.unsqueeze(1) * w_rows
     out = torch.zeros(B, O, device=w_rows.device, dtype=w_rows.dtype)
-    out.index_add_(0, batch.sample_ids(), contrib)
+    ops.scatter_add_rows(out, batch.sample_ids(), contrib)
     return out
 
 

@@ -90,3 +90,17 @@ def module(required: bool = False):
             "or multiverso_amd.ops.build(); GPU table ops refuse to run "
             "without the native gfx950 kernels.")
     return None
+
+
+def scatter_add_rows(dst, idx, src, alpha: float = 1.0):
+    """dst[idx[i], :] += alpha * src[i, :] with duplicate-index
+    accumulation. On GPU this dispatches to k_row_scatter_add (atomic,
+    thread-per-element): torch's index_add_ falls into its
+    indexFuncLargeIndex path for few-column rows, measured 4-5x slower
+    on gfx950 (profiles/round1_pmc.md). CPU falls back to index_add_."""
+    import torch
+    if dst.is_cuda and dst.dtype == torch.float32 and dst.dim() == 2:
+        m = module(required=True)
+        m.row_scatter_add(dst, idx, src.contiguous(), alpha)
+        return
+    dst.index_add_(0, idx, src if alpha == 1.0 else alpha * src)
