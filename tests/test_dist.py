@@ -141,3 +141,18 @@ def test_handler_init_protocol():
 
 def test_async_pipeline():
     run_dist(_async_pipeline, 2)
+
+
+# ---- world_size=4: the partition math and collective paths the driver
+# exercises at N=4/8 GPUs (uneven shards: 11 = 2+2+2+5, 9 rows = 2+2+2+3)
+
+def test_array_sync_oracle_ws4():
+    run_dist(_array_sync_oracle, 4)
+
+
+def test_array_uneven_shards_ws4():
+    run_dist(_array_uneven, 4)
+
+
+def test_matrix_whole_and_rows_ws4():
+    run_dist(_matrix_whole_and_rows, 4)
