@@ -59,6 +59,19 @@ def barrier() -> None:
     Zoo.get().barrier()
 
 
+def net_bind(rank: int, endpoint: str) -> bool:
+    """MV_NetBind (src/multiverso.cpp:58-62): declare this process's rank
+    for an explicit, launcher-free rendezvous (the reference's ZMQ Bind
+    deployment mode)."""
+    return Zoo.get().net_bind(rank, endpoint)
+
+
+def net_connect(ranks: List[int], endpoints: List[str]) -> bool:
+    """MV_NetConnect (src/multiverso.cpp:64-68): full rank->endpoint map;
+    rank 0's endpoint hosts the rendezvous store. Call before init()."""
+    return Zoo.get().net_connect(ranks, endpoints)
+
+
 def workers_num() -> int:
     return Zoo.get().num_workers
 
@@ -168,7 +181,8 @@ class MatrixTableHandler:
 
 
 __all__ = [
-    "init", "shutdown", "barrier", "workers_num", "servers_num", "worker_id",
+    "init", "shutdown", "barrier", "net_bind", "net_connect",
+    "workers_num", "servers_num", "worker_id",
     "server_id", "is_master_worker", "rank", "size", "aggregate",
     "ArrayTable", "MatrixTable", "SparseMatrixTable", "KVTable",
     "ArrayTableHandler", "MatrixTableHandler",

@@ -53,6 +53,31 @@ class Zoo:
         self._next_table_id = 0
         self._owns_pg = False
         self.backend = "none"
+        # explicit rendezvous (MV_NetBind/MV_NetConnect,
+        # src/multiverso.cpp:58-68): rank + rank->endpoint map set before
+        # init, replacing the launcher-env rendezvous the way the
+        # reference's ZMQ Bind/Connect replaced mpirun
+        self._bound_rank: Optional[int] = None
+        self._endpoints: Optional[List[str]] = None
+
+    # ---- explicit rendezvous (MV_NetBind / MV_NetConnect) ----
+    def net_bind(self, rank: int, endpoint: str) -> bool:
+        """Declare this process's rank (endpoint kept for the map;
+        reference ZMQNetWrapper::Bind, zmq_net.h)."""
+        CHECK(not self.started, "net_bind must precede init")
+        self._bound_rank = rank
+        return True
+
+    def net_connect(self, ranks: List[int], endpoints: List[str]) -> bool:
+        """Provide the full rank->endpoint map; rank 0's endpoint hosts
+        the rendezvous store (reference ZMQNetWrapper::Connect)."""
+        CHECK(not self.started, "net_connect must precede init")
+        CHECK(len(ranks) == len(endpoints), "ranks/endpoints mismatch")
+        eps = [""] * len(ranks)
+        for r, e in zip(ranks, endpoints):
+            eps[r] = e
+        self._endpoints = eps
+        return True
 
     # ---- singleton ----
     @classmethod
@@ -77,11 +102,20 @@ class Zoo:
               "MI355X rebuild runs every rank as worker+server (role=default); "
               f"got ps_role={role}")
 
-        world_size = int(os.environ.get("WORLD_SIZE", "1"))
+        explicit = (self._bound_rank is not None
+                    and self._endpoints is not None)
+        if explicit:
+            world_size = len(self._endpoints)
+        else:
+            world_size = int(os.environ.get("WORLD_SIZE", "1"))
         cuda = torch.cuda.is_available()
 
         if cuda:
-            local_rank = int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", "0")))
+            if explicit:
+                local_rank = self._bound_rank % torch.cuda.device_count()
+            else:
+                local_rank = int(os.environ.get(
+                    "LOCAL_RANK", os.environ.get("RANK", "0")))
             torch.cuda.set_device(local_rank)
             self.device = torch.device("cuda", local_rank)
         else:
@@ -90,6 +124,16 @@ class Zoo:
         if dist.is_initialized():
             self._owns_pg = False
             self.backend = dist.get_backend()
+        elif explicit and world_size > 1:
+            self.backend = backend or os.environ.get(
+                "MV_BACKEND", "nccl" if cuda else "gloo")
+            dist.init_process_group(
+                backend=self.backend,
+                init_method=f"tcp://{self._endpoints[0]}",
+                rank=self._bound_rank, world_size=world_size,
+                timeout=datetime.timedelta(seconds=300),
+            )
+            self._owns_pg = True
         elif world_size > 1 or "MASTER_ADDR" in os.environ:
             self.backend = backend or os.environ.get(
                 "MV_BACKEND", "nccl" if cuda else "gloo")

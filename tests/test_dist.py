@@ -143,6 +143,38 @@ def test_async_pipeline():
     run_dist(_async_pipeline, 2)
 
 
+def _net_bind_connect(rank, world, port):
+    # explicit rendezvous: no MASTER_ADDR/RANK/WORLD_SIZE in env
+    for k in ("MASTER_ADDR", "MASTER_PORT", "RANK", "WORLD_SIZE",
+              "LOCAL_RANK"):
+        import os
+        os.environ.pop(k, None)
+    import multiverso_amd as mv
+    eps = [f"127.0.0.1:{port}", f"127.0.0.1:{port + 1}"]
+    assert mv.net_bind(rank, eps[rank])
+    assert mv.net_connect([0, 1], eps)
+    mv.init(sync=True)
+    assert mv.size() == world and mv.rank() == rank
+    t = mv.ArrayTable(6)
+    t.add(torch.full((6,), 3.0)).wait()
+    assert torch.equal(t.get(), torch.full((6,), 3.0 * world))
+    mv.shutdown()
+
+
+def test_net_bind_connect():
+    """MV_NetBind/MV_NetConnect deployment mode (multiverso.cpp:58-68):
+    rendezvous from an explicit endpoint map, no launcher env."""
+    import torch.multiprocessing as mp
+    from conftest import free_port
+    port = free_port()
+    mp.start_processes(_net_bind_entry, args=(port,), nprocs=2,
+                       join=True, start_method="spawn")
+
+
+def _net_bind_entry(rank, port):
+    _net_bind_connect(rank, 2, port)
+
+
 # ---- world_size=4: the partition math and collective paths the driver
 # exercises at N=4/8 GPUs (uneven shards: 11 = 2+2+2+5, 9 rows = 2+2+2+3)
 
