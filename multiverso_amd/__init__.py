@@ -17,7 +17,6 @@ Public API parity with the reference Python binding
 
 from __future__ import annotations
 
-import sys
 from typing import List, Optional, Sequence
 
 import numpy as np

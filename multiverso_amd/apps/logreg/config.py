@@ -7,7 +7,7 @@ unknown keys warn and are ignored."""
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field, fields
+from dataclasses import dataclass, fields
 
 
 @dataclass

@@ -11,7 +11,6 @@ from __future__ import annotations
 import time
 from typing import Iterator, List, Optional
 
-import torch
 
 import multiverso_amd as mv
 

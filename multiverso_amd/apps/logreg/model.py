@@ -18,7 +18,7 @@ collectives. Worker-side lr decay matches SGDUpdater
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List
 
 import numpy as np
 import torch

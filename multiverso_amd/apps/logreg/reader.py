@@ -13,7 +13,7 @@ from __future__ import annotations
 import struct
 import threading
 from queue import Queue
-from typing import Iterator, List, Optional, Tuple
+from typing import Iterator, List, Tuple
 
 import torch
 

@@ -22,8 +22,7 @@ MI355X-native redesign of the training loop:
 
 from __future__ import annotations
 
-import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional, Tuple
 
 import torch

@@ -11,7 +11,6 @@ MULTIVERSO_USE_HDFS."""
 
 from __future__ import annotations
 
-import os
 from typing import Callable, Dict, Optional
 
 

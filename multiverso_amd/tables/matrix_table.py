@@ -21,7 +21,7 @@ bitmap) is implemented in sparse_matrix.py on top of this class.
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import Optional, Tuple
 
 import torch
 

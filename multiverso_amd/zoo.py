@@ -27,7 +27,7 @@ from typing import Dict, List, Optional
 import torch
 import torch.distributed as dist
 
-from .configure import get_flag, parse_cmd_flags, set_flag
+from .configure import get_flag, parse_cmd_flags
 from .log import CHECK, log
 
 
