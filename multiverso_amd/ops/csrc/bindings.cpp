@@ -16,6 +16,8 @@ void mv_launch_sgd(float*, const float*, int64_t, hipStream_t);
 void mv_launch_momentum(float*, float*, const float*, float, int64_t, hipStream_t);
 void mv_launch_adagrad(float*, float*, const float*, float, float, float,
                        int64_t, hipStream_t);
+void mv_launch_sgd_copy(float*, const float*, float*, float, int64_t,
+                        hipStream_t);
 void mv_launch_dcasgd(float*, float*, const float*, float, float, int64_t,
                       hipStream_t);
 void mv_launch_dcasgda(float*, float*, float*, const float*, float, float,
@@ -87,6 +89,16 @@ void adagrad_update(torch::Tensor data, torch::Tensor gsq, torch::Tensor delta,
   mv_launch_adagrad(data.data_ptr<float>(), gsq.data_ptr<float>(),
                     delta.data_ptr<float>(), (float)lr, (float)rho, (float)eps,
                     data.numel(), cur_stream());
+}
+
+void sgd_copy_update(torch::Tensor data, torch::Tensor delta,
+                     torch::Tensor out, double sign) {
+  check_f32(data, "data"); check_f32(delta, "delta"); check_f32(out, "out");
+  TORCH_CHECK(data.numel() == delta.numel() && data.numel() == out.numel(),
+              "size mismatch");
+  mv_launch_sgd_copy(data.data_ptr<float>(), delta.data_ptr<float>(),
+                     out.data_ptr<float>(), (float)sign, data.numel(),
+                     cur_stream());
 }
 
 void dcasgd_update(torch::Tensor data, torch::Tensor bak, torch::Tensor delta,
@@ -235,6 +247,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_update", &sgd_update, "K2: data -= delta");
   m.def("momentum_update", &momentum_update, "K3: fused momentum update");
   m.def("adagrad_update", &adagrad_update, "K4: fused adagrad update");
+  m.def("sgd_copy_update", &sgd_copy_update,
+        "fused Add+Get: data (+/-)= delta; out = data (saves the Get's "
+        "shard re-read at N=1)");
   m.def("dcasgd_update", &dcasgd_update,
         "DC-ASGD: delay-compensated update with per-worker backup");
   m.def("dcasgda_update", &dcasgda_update,

@@ -136,6 +136,35 @@ __global__ void k_adagrad_tail(float* __restrict__ data, float* __restrict__ gsq
   }
 }
 
+// Fused Add+Get (single-rank fast path): an Add immediately followed by
+// a whole-table Get re-reads the shard the updater just wrote. Fusing
+// the Get's copy-out into the updater pass saves that re-read — 2.0 GB
+// instead of 2.5 GB per Add+Get step on the 1e6x128 headline config
+// (observable semantics identical: shard updated AND out filled with the
+// updated values).
+__global__ void k_sgd_copy_f4(v4f* __restrict__ data,
+                              const v4f* __restrict__ delta,
+                              v4f* __restrict__ out, float sign, int64_t n4) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    v4f v = ntload(&data[i]) + sign * ntload(&delta[i]);
+    ntstore(&data[i], v);
+    ntstore(&out[i], v);
+  }
+}
+
+__global__ void k_sgd_copy_tail(float* __restrict__ data,
+                                const float* __restrict__ delta,
+                                float* __restrict__ out, float sign,
+                                int64_t start, int64_t n) {
+  int64_t i = start + blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    float v = data[i] + sign * delta[i];
+    data[i] = v;
+    out[i] = v;
+  }
+}
+
 // DC-ASGD updaters ("dcasgd"/"dcasgda", selected at reference
 // updater.cpp:51-54 from a submodule ABSENT from the snapshot
 // (.gitmodules:1-3, empty dir) — math reconstructed from the DC-ASGD
@@ -321,6 +350,17 @@ void mv_launch_adagrad(float* data, float* gsq, const float* delta,
   int64_t tail = n - n4 * 4;
   if (tail) k_adagrad_tail<<<1, 64, 0, s>>>(data, gsq, delta, inv_lr, rho, eps,
                                             n4 * 4, n);
+}
+
+// sign = +1 for the default (add) updater, -1 for sgd.
+void mv_launch_sgd_copy(float* data, const float* delta, float* out,
+                        float sign, int64_t n, hipStream_t s) {
+  int64_t n4 = n / 4;
+  if (n4) k_sgd_copy_f4<<<grid_for_cap(n4, ELEM_GRID), BLOCK, 0, s>>>(
+      (v4f*)data, (const v4f*)delta, (v4f*)out, sign, n4);
+  int64_t tail = n - n4 * 4;
+  if (tail) k_sgd_copy_tail<<<1, 64, 0, s>>>(data, delta, out, sign,
+                                             n4 * 4, n);
 }
 
 void mv_launch_dcasgd(float* data, float* bak, const float* delta,

@@ -57,9 +57,49 @@ class MatrixTable(Table):
             self.shard.copy_(full[self.row_offset:
                                   self.row_offset + self.local_rows])
         self._make_updater(self.shard.view(-1))
+        # single-rank Add-deferral for stateless updaters: an Add
+        # immediately followed by a whole-table Get fuses into one kernel
+        # (k_sgd_copy) that writes both the shard and the Get buffer,
+        # saving the Get's shard re-read (0.5 GB/step on the headline
+        # config). Any other table op materializes the Add first (flush).
+        self._deferred = None  # (delta, option, delta._version)
+
+    def _check_deferred(self, d) -> None:
+        CHECK(d[0]._version == d[2],
+              "the delta tensor passed to Add was mutated in place before "
+              "the Add was applied (single-GPU deferred-Add fast path "
+              "snapshots at the next table op; pass a fresh tensor)")
+
+    def flush(self) -> None:
+        d = self._deferred
+        if d is not None:
+            self._deferred = None
+            self._check_deferred(d)
+            with monitor("server.update"):
+                self.updater.update(d[0], d[1])
+        super().flush()
 
     # ---- whole-table ops ----
     def get(self, out: Optional[torch.Tensor] = None, async_op: bool = False):
+        d = self._deferred
+        if d is not None and not async_op:
+            self._deferred = None
+            self._check_deferred(d)
+            if out is None:
+                out = torch.empty(self.num_row, self.num_col,
+                                  dtype=self.dtype, device=self.device)
+            CHECK(out.numel() == self.num_row * self.num_col,
+                  "Get buffer size mismatch")
+            if out.is_contiguous():
+                from .. import ops
+                sign = -1.0 if self.updater_type == "sgd" else 1.0
+                with monitor("server.update"):
+                    ops.module(required=True).sgd_copy_update(
+                        self.shard.view(-1), d[0], out.view(-1), sign)
+                return out
+            # non-contiguous Get buffer: materialize, fall through
+            with monitor("server.update"):
+                self.updater.update(d[0], d[1])
         self.flush()
         if out is None:
             out = torch.empty(self.num_row, self.num_col, dtype=self.dtype,
@@ -79,6 +119,11 @@ class MatrixTable(Table):
         CHECK(delta.numel() == self.num_row * self.num_col,
               "Add delta size mismatch")
         delta = delta.to(self.device, self.dtype).contiguous().view(-1)
+        if (self.zoo.size == 1 and self.shard.is_cuda
+                and self.updater_type in ("sgd", "default")):
+            self.flush()                 # at most one deferred Add
+            self._deferred = (delta, option, delta._version)
+            return Handle()
         with monitor("worker.add"):
             chunk, h = reduce_scatter_delta(delta, self.spec, self.num_col,
                                             async_op=async_op)
@@ -178,6 +223,7 @@ class MatrixTable(Table):
     def add_rows(self, row_ids, values: torch.Tensor,
                  option: Optional[AddOption] = None) -> None:
         """Row-subset Add (matrix_table.cpp:265-309 partition path)."""
+        self.flush()   # a deferred whole-table Add must land first
         ids = torch.as_tensor(row_ids, dtype=torch.int64, device=self.device)
         vals = values.to(self.device, self.dtype).contiguous()
         CHECK(vals.numel() == ids.numel() * self.num_col,
@@ -199,6 +245,7 @@ class MatrixTable(Table):
         self.zoo.barrier()
 
     def load(self, path: str) -> None:
+        self.flush()
         import numpy as np
         arr = np.fromfile(path, dtype=str(self.dtype).replace("torch.", ""))
         CHECK(arr.size == self.num_row * self.num_col,

@@ -56,6 +56,7 @@ class SparseMatrixTable(MatrixTable):
 
     def add_rows(self, row_ids, values, option: Optional[AddOption] = None,
                  source_worker: Optional[int] = None) -> None:
+        self.flush()   # a deferred whole-table Add must land first
         ids = torch.as_tensor(row_ids, dtype=torch.int64, device=self.device)
         vals = values.to(self.device, self.dtype).contiguous()
         from ..comm import all_to_all_rows

@@ -132,3 +132,46 @@ def test_native_extension_is_mandatory_on_gpu():
     from multiverso_amd import ops
     m = ops.module(required=True)
     assert m.__file__.endswith("_mv_hip.so")
+
+
+def test_fused_add_get_matches_sequential():
+    """Single-rank deferred Add + Get fuses into k_sgd_copy; results must
+    equal the sequential add-then-copy semantics, and ordering with
+    row ops / a second add must be preserved by the flush points."""
+    import multiverso_amd as mv
+    mv.init()
+    t = mv.MatrixTable(1000, 32, updater_type="sgd")
+    delta = torch.randn(1000, 32, device="cuda:0")
+    got = None
+    t.add(delta)
+    got = t.get()
+    torch.cuda.synchronize()
+    assert torch.allclose(got, -delta, rtol=1e-6, atol=1e-7)
+    assert torch.allclose(t.shard, -delta, rtol=1e-6, atol=1e-7)
+    # deferred add followed by add_rows: the whole-table add lands first
+    t.add(delta)
+    t.add_rows([5], torch.full((1, 32), 7.0, device="cuda:0"))
+    rows = t.get_rows([5, 6])
+    torch.cuda.synchronize()
+    assert torch.allclose(rows[0], -2 * delta[5] - 7.0, rtol=1e-5, atol=1e-6)
+    assert torch.allclose(rows[1], -2 * delta[6], rtol=1e-5, atol=1e-6)
+    # two back-to-back adds: first materializes when the second defers
+    t2 = mv.MatrixTable(64, 16, updater_type="default")
+    one = torch.ones(64, 16, device="cuda:0")
+    t2.add(one)
+    t2.add(one)
+    out = t2.get()
+    torch.cuda.synchronize()
+    assert torch.equal(out, 2 * one)
+    # in-place mutation of the delta between Add and Get is a loud error
+    from multiverso_amd.log import FatalError
+    t3 = mv.MatrixTable(64, 16, updater_type="sgd")
+    d3 = torch.ones(64, 16, device="cuda:0")
+    t3.add(d3)
+    d3.mul_(2.0)
+    try:
+        t3.get()
+        assert False, "expected FatalError on mutated deferred delta"
+    except FatalError:
+        pass
+    mv.shutdown()
