@@ -353,10 +353,13 @@ void mv_launch_adagrad(float* data, float* gsq, const float* delta,
 }
 
 // sign = +1 for the default (add) updater, -1 for sgd.
+// Grid 512 (2 blocks/CU): the 2-read/2-write stream measured 6.0 TB/s at
+// 512 blocks vs 5.35 at 1024+ (tools/probe_sgd.hip fused sweep) — the
+// wider grid over-subscribes the write queues.
 void mv_launch_sgd_copy(float* data, const float* delta, float* out,
                         float sign, int64_t n, hipStream_t s) {
   int64_t n4 = n / 4;
-  if (n4) k_sgd_copy_f4<<<grid_for_cap(n4, ELEM_GRID), BLOCK, 0, s>>>(
+  if (n4) k_sgd_copy_f4<<<grid_for_cap(n4, 512), BLOCK, 0, s>>>(
       (v4f*)data, (const v4f*)delta, (v4f*)out, sign, n4);
   int64_t tail = n - n4 * 4;
   if (tail) k_sgd_copy_tail<<<1, 64, 0, s>>>(data, delta, out, sign,

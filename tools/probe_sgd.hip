@@ -37,6 +37,30 @@ __global__ void sgd_f4_ntboth(float4* __restrict__ d, const float4* __restrict__
   }
 }
 
+// fused Add+Get: read d, read g, write d, write o = 16 B/element
+__global__ void sgd_copy_nt(float4* __restrict__ d, const float4* __restrict__ g,
+                            float4* __restrict__ o, long n4) {
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    v4f a = __builtin_nontemporal_load((v4f*)&d[i]);
+    v4f b = __builtin_nontemporal_load((const v4f*)&g[i]);
+    a -= b;
+    __builtin_nontemporal_store(a, (v4f*)&d[i]);
+    __builtin_nontemporal_store(a, (v4f*)&o[i]);
+  }
+}
+
+__global__ void sgd_copy_plain(float4* __restrict__ d, const float4* __restrict__ g,
+                               float4* __restrict__ o, long n4) {
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    float4 a = d[i], b = g[i];
+    a.x -= b.x; a.y -= b.y; a.z -= b.z; a.w -= b.w;
+    d[i] = a;
+    o[i] = a;
+  }
+}
+
 int main() {
   long n = 128L * 1000 * 1000;  // 1e6x128
   long n4 = n / 4;
@@ -65,6 +89,28 @@ int main() {
     bench(sgd_f4, grid, "plain  ");
     bench(sgd_f4_nt, grid, "nt-load");
     bench(sgd_f4_ntboth, grid, "nt-both");
+  }
+  float* o;
+  (void)hipMalloc(&o, n * 4);
+  auto bench3 = [&](auto kern, int grid, const char* name) {
+    kern<<<grid, BLOCK>>>((float4*)d, (const float4*)g, (float4*)o, n4);
+    (void)hipDeviceSynchronize();
+    hipEvent_t a, b;
+    (void)hipEventCreate(&a); (void)hipEventCreate(&b);
+    (void)hipEventRecord(a);
+    for (int r = 0; r < 5; ++r)
+      kern<<<grid, BLOCK>>>((float4*)d, (const float4*)g, (float4*)o, n4);
+    (void)hipEventRecord(b);
+    (void)hipEventSynchronize(b);
+    float ms;
+    (void)hipEventElapsedTime(&ms, a, b);
+    ms /= 5;
+    printf("%s grid=%5d: %.3f ms  %.2f TB/s\n", name, grid, ms,
+           n * 16.0 / (ms * 1e-3) / 1e12);
+  };
+  for (int grid : {512, 1024, 2048, 4096, 8192}) {
+    bench3(sgd_copy_nt, grid, "fused-nt   ");
+    bench3(sgd_copy_plain, grid, "fused-plain");
   }
   return 0;
 }
