@@ -25,8 +25,11 @@
 #define MAX_GRID 2048
 // elementwise streaming kernels saturate HBM at a smaller grid
 // (tools/probe_sgd.hip sweep: 5.85 TB/s at 1024 blocks w/ nontemporal
-// access vs 5.16 at 2048 plain)
+// access vs 5.16 at 2048 plain). Kernels with >=4 concurrent HBM streams
+// (fused add+get, momentum/adagrad/dcasgd state updates) peak lower
+// still: 6.05/5.84 TB/s at 768 blocks vs 5.45/5.15 at 1024 (same sweep).
 #define ELEM_GRID 1024
+#define STATE_GRID 768
 
 typedef float v4f __attribute__((ext_vector_type(4)));
 
@@ -335,7 +338,7 @@ void mv_launch_sgd(float* data, const float* delta, int64_t n, hipStream_t s) {
 void mv_launch_momentum(float* data, float* m, const float* delta, float mu,
                         int64_t n, hipStream_t s) {
   int64_t n4 = n / 4;
-  if (n4) k_momentum_f4<<<grid_for_cap(n4, ELEM_GRID), BLOCK, 0, s>>>(
+  if (n4) k_momentum_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
       (v4f*)data, (v4f*)m, (const v4f*)delta, mu, n4);
   int64_t tail = n - n4 * 4;
   if (tail) k_momentum_tail<<<1, 64, 0, s>>>(data, m, delta, mu, n4 * 4, n);
@@ -345,7 +348,7 @@ void mv_launch_adagrad(float* data, float* gsq, const float* delta,
                        float lr, float rho, float eps, int64_t n, hipStream_t s) {
   float inv_lr = 1.0f / lr;
   int64_t n4 = n / 4;
-  if (n4) k_adagrad_f4<<<grid_for_cap(n4, ELEM_GRID), BLOCK, 0, s>>>(
+  if (n4) k_adagrad_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
       (v4f*)data, (v4f*)gsq, (const v4f*)delta, inv_lr, rho, eps, n4);
   int64_t tail = n - n4 * 4;
   if (tail) k_adagrad_tail<<<1, 64, 0, s>>>(data, gsq, delta, inv_lr, rho, eps,
@@ -353,13 +356,13 @@ void mv_launch_adagrad(float* data, float* gsq, const float* delta,
 }
 
 // sign = +1 for the default (add) updater, -1 for sgd.
-// Grid 512 (2 blocks/CU): the 2-read/2-write stream measured 6.0 TB/s at
-// 512 blocks vs 5.35 at 1024+ (tools/probe_sgd.hip fused sweep) — the
+// Grid 768 (3 blocks/CU): the 2-read/2-write stream measured 6.05 TB/s
+// at 768 blocks vs 5.45 at 1024+ (tools/probe_sgd.hip fused sweep) — a
 // wider grid over-subscribes the write queues.
 void mv_launch_sgd_copy(float* data, const float* delta, float* out,
                         float sign, int64_t n, hipStream_t s) {
   int64_t n4 = n / 4;
-  if (n4) k_sgd_copy_f4<<<grid_for_cap(n4, 512), BLOCK, 0, s>>>(
+  if (n4) k_sgd_copy_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
       (v4f*)data, (const v4f*)delta, (v4f*)out, sign, n4);
   int64_t tail = n - n4 * 4;
   if (tail) k_sgd_copy_tail<<<1, 64, 0, s>>>(data, delta, out, sign,
@@ -369,7 +372,7 @@ void mv_launch_sgd_copy(float* data, const float* delta, float* out,
 void mv_launch_dcasgd(float* data, float* bak, const float* delta,
                       float lr, float lambda, int64_t n, hipStream_t s) {
   int64_t n4 = n / 4;
-  if (n4) k_dcasgd_f4<<<grid_for_cap(n4, ELEM_GRID), BLOCK, 0, s>>>(
+  if (n4) k_dcasgd_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
       (v4f*)data, (v4f*)bak, (const v4f*)delta, lr, lambda, n4);
   int64_t tail = n - n4 * 4;
   if (tail) k_dcasgd_tail<<<1, 64, 0, s>>>(data, bak, delta, lr, lambda,
@@ -380,7 +383,7 @@ void mv_launch_dcasgda(float* data, float* bak, float* msq, const float* delta,
                        float lr, float lambda, float rho, float eps,
                        int64_t n, hipStream_t s) {
   int64_t n4 = n / 4;
-  if (n4) k_dcasgda_f4<<<grid_for_cap(n4, ELEM_GRID), BLOCK, 0, s>>>(
+  if (n4) k_dcasgda_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
       (v4f*)data, (v4f*)bak, (v4f*)msq, (const v4f*)delta, lr, lambda, rho,
       eps, n4);
   int64_t tail = n - n4 * 4;

@@ -61,6 +61,22 @@ __global__ void sgd_copy_plain(float4* __restrict__ d, const float4* __restrict_
   }
 }
 
+// momentum/adagrad shape: 3 reads (data, state, delta) + 2 writes
+// (data, state) = 20 B/element
+__global__ void mom_nt(float4* __restrict__ d, float4* __restrict__ m,
+                       const float4* __restrict__ g, long n4) {
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    v4f a = __builtin_nontemporal_load((v4f*)&d[i]);
+    v4f mm = __builtin_nontemporal_load((v4f*)&m[i]);
+    v4f b = __builtin_nontemporal_load((const v4f*)&g[i]);
+    mm = 0.9f * mm + 0.1f * b;
+    a -= mm;
+    __builtin_nontemporal_store(mm, (v4f*)&m[i]);
+    __builtin_nontemporal_store(a, (v4f*)&d[i]);
+  }
+}
+
 int main() {
   long n = 128L * 1000 * 1000;  // 1e6x128
   long n4 = n / 4;
@@ -108,9 +124,26 @@ int main() {
     printf("%s grid=%5d: %.3f ms  %.2f TB/s\n", name, grid, ms,
            n * 16.0 / (ms * 1e-3) / 1e12);
   };
-  for (int grid : {512, 1024, 2048, 4096, 8192}) {
+  for (int grid : {256, 384, 512, 768, 1024, 2048}) {
     bench3(sgd_copy_nt, grid, "fused-nt   ");
-    bench3(sgd_copy_plain, grid, "fused-plain");
   }
+  // momentum shape (m reuses o as state; 20 B/element)
+  auto bench_m = [&](int grid) {
+    mom_nt<<<grid, BLOCK>>>((float4*)d, (float4*)o, (const float4*)g, n4);
+    (void)hipDeviceSynchronize();
+    hipEvent_t a, b;
+    (void)hipEventCreate(&a); (void)hipEventCreate(&b);
+    (void)hipEventRecord(a);
+    for (int r = 0; r < 5; ++r)
+      mom_nt<<<grid, BLOCK>>>((float4*)d, (float4*)o, (const float4*)g, n4);
+    (void)hipEventRecord(b);
+    (void)hipEventSynchronize(b);
+    float ms;
+    (void)hipEventElapsedTime(&ms, a, b);
+    ms /= 5;
+    printf("mom-nt grid=%5d: %.3f ms  %.2f TB/s\n", grid, ms,
+           n * 20.0 / (ms * 1e-3) / 1e12);
+  };
+  for (int grid : {256, 384, 512, 768, 1024, 2048}) bench_m(grid);
   return 0;
 }
