@@ -59,8 +59,9 @@ def _scores(batch: Batch, w_rows: torch.Tensor) -> torch.Tensor:
     no atomics (measured 3x the throughput of the scatter-add form on
     gfx950; fp64 accumulator keeps the diff exact for fp32 data)."""
     contrib = (batch.vals.unsqueeze(1) * w_rows).to(torch.float64)
-    cs = torch.zeros(contrib.size(0) + 1, contrib.size(1),
+    cs = torch.empty(contrib.size(0) + 1, contrib.size(1),
                      device=contrib.device, dtype=torch.float64)
+    cs[0].zero_()
     torch.cumsum(contrib, 0, out=cs[1:])
     ptr = batch.ptr.long()
     return (cs[ptr[1:]] - cs[ptr[:-1]]).to(w_rows.dtype)
