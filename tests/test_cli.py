@@ -1,0 +1,78 @@
+"""End-to-end CLI tests for the two applications (the reference's
+distributed_word_embedding and LogisticRegression binaries,
+SURVEY.md §2.11): corpus/config in, trained artifacts out."""
+
+import random
+import subprocess
+import sys
+import os
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_wordembedding_cli(tmp_path):
+    corpus = tmp_path / "corpus.txt"
+    out = tmp_path / "emb.txt"
+    rng = random.Random(3)
+    words = [f"w{i}" for i in range(100)]
+    with open(corpus, "w") as f:
+        for _ in range(400):
+            f.write(" ".join(rng.choices(words, k=10)) + "\n")
+    r = subprocess.run(
+        [sys.executable, "-m", "multiverso_amd.apps.wordembedding.main",
+         "-train_file", str(corpus), "-output", str(out), "-size", "16",
+         "-epoch", "1", "-min_count", "1", "-negative", "2"],
+        cwd=ROOT, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    lines = open(out).read().strip().split("\n")
+    vocab, dim = map(int, lines[0].split())
+    assert vocab == 100 and dim == 16
+    assert len(lines) == vocab + 1
+    first = lines[1].split()
+    assert len(first) == dim + 1 and first[0].startswith("w")
+
+
+def test_wordembedding_cli_binary_output(tmp_path):
+    corpus = tmp_path / "c.txt"
+    out = tmp_path / "emb.bin"
+    with open(corpus, "w") as f:
+        for _ in range(50):
+            f.write("a b c d e f g h\n")
+    r = subprocess.run(
+        [sys.executable, "-m", "multiverso_amd.apps.wordembedding.main",
+         "-train_file", str(corpus), "-output", str(out), "-size", "8",
+         "-epoch", "1", "-min_count", "1", "-negative", "2", "-binary", "1"],
+        cwd=ROOT, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    blob = open(out, "rb").read()
+    header = blob.split(b"\n", 1)[0]
+    vocab, dim = map(int, header.split())
+    assert vocab == 8 and dim == 8
+
+
+def test_logreg_cli(tmp_path):
+    train = tmp_path / "train.txt"
+    model = tmp_path / "model.bin"
+    rng = random.Random(5)
+    with open(train, "w") as f:
+        for _ in range(400):
+            keys = rng.sample(range(500), 10)
+            label = 1 if sum(keys) > 2500 else 0
+            f.write(f"{label} " + " ".join(f"{k}:1" for k in sorted(keys))
+                    + "\n")
+    cfg = tmp_path / "t.config"
+    cfg.write_text(
+        "input_size=500\noutput_size=1\nobjective_type=sigmoid\n"
+        "updater_type=sgd\nlearning_rate=0.1\nminibatch_size=32\n"
+        "train_epoch=2\nreader_type=default\nuse_ps=true\n"
+        f"sync_frequency=4\ntrain_file={train}\ntest_file={train}\n"
+        f"output_model_file={model}\n")
+    r = subprocess.run(
+        [sys.executable, "-m", "multiverso_amd.apps.logreg.main", str(cfg)],
+        cwd=ROOT, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    assert model.exists() and model.stat().st_size == 500 * 4
+    # training must beat chance on this separable-ish problem
+    import re
+    m = re.search(r"\((0\.\d+)\)", r.stdout + r.stderr)
+    assert m and float(m.group(1)) > 0.55, (r.stdout + r.stderr)[-400:]
