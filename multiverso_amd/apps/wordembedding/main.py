@@ -77,12 +77,37 @@ def main(argv=None) -> None:
         data_block_size=args.data_block_size)
     model = WordEmbedding(opt, dictionary.counts)
 
+    def block_stream(reader):
+        """is_pipeline=1 (default, reference
+        distributed_wordembedding.cpp:202-223): a loader thread tokenizes
+        and stages block N+1 while block N trains on the GPU."""
+        if not args.is_pipeline:
+            yield from reader.blocks()
+            return
+        import queue
+        import threading
+        q: "queue.Queue" = queue.Queue(maxsize=2)
+
+        def loader():
+            for blk in reader.blocks():
+                q.put(blk)
+            q.put(None)
+
+        t = threading.Thread(target=loader, daemon=True)
+        t.start()
+        while True:
+            blk = q.get()
+            if blk is None:
+                break
+            yield blk
+        t.join()
+
     t0 = time.perf_counter()
     trained = 0
     for epoch in range(args.epoch):
         reader = TextBlockReader(args.train_file, dictionary,
                                  args.data_block_size, mv.rank(), mv.size())
-        for words, sids in reader.blocks():
+        for words, sids in block_stream(reader):
             trained += model.train_block(words, sids)
             model.sync_word_count()
             if mv.rank() == 0:
