@@ -313,8 +313,9 @@ class WordEmbedding:
             if ns_fast:
                 self._train_kernel_ns(self.input_table.shard,
                                       self.output_table.shard, igq, ogq,
-                                      in_idx, in_off, centers,
-                                      self._block_pool)
+                                      in_idx,
+                                      in_off if opt.cbow else None,
+                                      centers, self._block_pool)
             else:
                 self._train_kernel(self.input_table.shard,
                                    self.output_table.shard, igq, ogq,
@@ -357,7 +358,9 @@ class WordEmbedding:
 
         if ns_fast:
             self._train_kernel_ns(in_buf, out_buf, gbufs[0], gbufs[2],
-                                  in_local, in_off, out_local, pool_local)
+                                  in_local,
+                                  in_off if opt.cbow else None,
+                                  out_local, pool_local)
         else:
             self._train_kernel(in_buf, out_buf, gbufs[0], gbufs[2],
                                in_local, in_off, out_local,
@@ -387,14 +390,17 @@ class WordEmbedding:
 
     def _train_kernel_ns(self, in_buf, out_buf, in_gsq, out_gsq,
                          in_local, in_off, centers, pool):
-        """NS fast path: negatives generated in-kernel from `pool`."""
+        """NS fast path: negatives generated in-kernel from `pool`.
+        ``in_off=None`` = skip-gram (exactly one input per group; the
+        kernel skips the offset loads — probe V7)."""
         from ... import ops
         hip = ops.module(required=True)
         dummy = in_buf
         igq = in_gsq if in_gsq is not None else dummy
         ogq = out_gsq if out_gsq is not None else dummy
-        in_off = in_off.to(torch.int32)
-        g_total = in_off.numel() - 1
+        if in_off is not None:
+            in_off = in_off.to(torch.int32)
+        g_total = centers.numel()
         step = self.opt.max_groups_per_launch
         if g_total <= step:
             hip.w2v_train_ns(in_buf, out_buf, igq, ogq, in_local, in_off,
@@ -403,12 +409,13 @@ class WordEmbedding:
                              self.opt.use_adagrad, self.opt.atomic_updates)
             return
         gs = list(range(0, g_total, step)) + [g_total]
-        ibs = in_off[gs].tolist()
+        ibs = gs if in_off is None else in_off[gs].tolist()
         for k in range(len(gs) - 1):
             g0, g1 = gs[k], gs[k + 1]
             i0, i1 = ibs[k], ibs[k + 1]
             hip.w2v_train_ns(in_buf, out_buf, igq, ogq,
                              in_local[i0:i1].contiguous(),
+                             None if in_off is None else
                              (in_off[g0:g1 + 1] - i0).contiguous(),
                              centers[g0:g1].contiguous(), pool,
                              self.opt.negative_num, self._next_seed(),

@@ -208,7 +208,8 @@ void w2v_train(torch::Tensor in_emb, torch::Tensor out_emb,
 
 void w2v_train_ns(torch::Tensor in_emb, torch::Tensor out_emb,
                   torch::Tensor in_gsq, torch::Tensor out_gsq,
-                  torch::Tensor in_idx, torch::Tensor in_off,
+                  torch::Tensor in_idx,
+                  c10::optional<torch::Tensor> in_off_opt,
                   torch::Tensor centers, torch::Tensor pool, int64_t neg,
                   int64_t seed, double lr, bool use_adagrad,
                   bool use_atomic) {
@@ -219,10 +220,20 @@ void w2v_train_ns(torch::Tensor in_emb, torch::Tensor out_emb,
   TORCH_CHECK(in_idx.scalar_type() == torch::kInt64 &&
               centers.scalar_type() == torch::kInt64 &&
               pool.scalar_type() == torch::kInt64, "ids must be int64");
-  TORCH_CHECK(in_off.scalar_type() == torch::kInt32, "offsets must be int32");
+  const int* in_off_ptr = nullptr;
+  if (in_off_opt.has_value()) {
+    TORCH_CHECK(in_off_opt->scalar_type() == torch::kInt32,
+                "offsets must be int32");
+    TORCH_CHECK(in_off_opt->numel() - 1 == centers.numel(),
+                "centers/group count mismatch");
+    in_off_ptr = in_off_opt->data_ptr<int>();
+  } else {
+    // skip-gram: one input per group (in_idx[g] is group g's context)
+    TORCH_CHECK(in_idx.numel() == centers.numel(),
+                "skip-gram in_idx/centers mismatch");
+  }
   TORCH_CHECK(pool.numel() > 0, "empty negative pool");
-  int64_t G = in_off.numel() - 1;
-  TORCH_CHECK(centers.numel() == G, "centers/group count mismatch");
+  int64_t G = centers.numel();
   float *igq = nullptr, *ogq = nullptr;
   if (use_adagrad) {
     check_f32(in_gsq, "in_gsq"); check_f32(out_gsq, "out_gsq");
@@ -232,7 +243,7 @@ void w2v_train_ns(torch::Tensor in_emb, torch::Tensor out_emb,
     ogq = out_gsq.data_ptr<float>();
   }
   mv_launch_w2v_ns(in_emb.data_ptr<float>(), out_emb.data_ptr<float>(), igq,
-                   ogq, in_idx.data_ptr<int64_t>(), in_off.data_ptr<int>(),
+                   ogq, in_idx.data_ptr<int64_t>(), in_off_ptr,
                    centers.data_ptr<int64_t>(), pool.data_ptr<int64_t>(),
                    pool.numel(), (int)neg, (uint64_t)seed, (float)lr, G,
                    in_emb.size(1), use_adagrad ? 1 : 0, use_atomic ? 1 : 0,

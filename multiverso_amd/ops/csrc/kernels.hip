@@ -549,7 +549,9 @@ __global__ void k_w2v_ns(float* __restrict__ in_emb, float* __restrict__ out_emb
     float h[DPL], err[DPL];
 #pragma unroll
     for (int d = 0; d < DPL; ++d) { h[d] = 0.f; err[d] = 0.f; }
-    int ib = in_off[g], ie = in_off[g + 1];
+    // in_off == nullptr: skip-gram (exactly one input per group) — the
+    // offset loads vanish (probe V7: 5% over the ragged form)
+    int ib = in_off ? in_off[g] : g, ie = in_off ? in_off[g + 1] : g + 1;
     for (int i = ib; i < ie; ++i) {
       const float* row = in_emb + in_idx[i] * dim;
 #pragma unroll

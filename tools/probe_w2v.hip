@@ -247,6 +247,79 @@ __global__ void k_w2v_probe_pf(float* __restrict__ in_emb,
   }
 }
 
+// VARIANT 6: mirror of the production k_w2v_ns — in-kernel LCG negatives
+// from a pool, plain stores; isolates the pool-indirection cost vs V1.
+// VARIANT 7: V6 + skip-gram specialization (1 input/group, no in_off).
+template <int DPL, bool NIN1>
+__global__ void k_w2v_probe_ns(float* __restrict__ in_emb,
+                               float* __restrict__ out_emb,
+                               const long* __restrict__ in_idx,
+                               const int* __restrict__ in_off,
+                               const long* __restrict__ centers,
+                               const long* __restrict__ pool, long pool_n,
+                               int neg, unsigned long long seed,
+                               float lr, int G, int dim) {
+  int wid = (int)((blockIdx.x * (long)blockDim.x + threadIdx.x) >> 6);
+  int lane = threadIdx.x & 63;
+  int nwaves = (int)((gridDim.x * (long)blockDim.x) >> 6);
+  for (int g = wid; g < G; g += nwaves) {
+    float h[DPL], err[DPL];
+#pragma unroll
+    for (int d = 0; d < DPL; ++d) { h[d] = 0.f; err[d] = 0.f; }
+    int ib = NIN1 ? g : in_off[g], ie = NIN1 ? g + 1 : in_off[g + 1];
+    for (int i = ib; i < ie; ++i) {
+      const float* row = in_emb + in_idx[i] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) h[d] += row[c];
+      }
+    }
+    long pos = centers[g];
+    unsigned long long next_random = seed + (unsigned long long)g * 25214903917ull + 11ull;
+    for (int o = 0; o <= neg; ++o) {
+      long node;
+      float label;
+      if (o == 0) { node = pos; label = 1.f; }
+      else {
+        next_random = next_random * 25214903917ull + 11ull;
+        node = pool[(long)((next_random >> 8) % (unsigned long long)pool_n)];
+        label = 0.f;
+        if (node == pos) continue;
+      }
+      float* w = out_emb + node * dim;
+      float wv[DPL];
+      float f = 0.f;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        wv[d] = (c < dim) ? w[c] : 0.f;
+        f += h[d] * wv[d];
+      }
+#pragma unroll
+      for (int s = 32; s; s >>= 1) f += __shfl_xor(f, s, 64);
+      f = 1.f / (1.f + expf(-f));
+      float e = label - f;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) {
+          err[d] += e * wv[d];
+          w[c] = wv[d] + e * lr * h[d];
+        }
+      }
+    }
+    for (int i = ib; i < ie; ++i) {
+      float* row = in_emb + in_idx[i] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) row[c] += lr * err[d];
+      }
+    }
+  }
+}
+
 int main(int argc, char** argv) {
   int G = argc > 1 ? atoi(argv[1]) : 1 << 21;
   int V = argc > 2 ? atoi(argv[2]) : 1000000;  // vocab
@@ -319,5 +392,34 @@ int main(int argc, char** argv) {
   bench(k_w2v_probe<4, 3>, "V3 read-only    ");
   bench(k_w2v_probe_pf<4, 8>, "V4 prefetch-8   ");
   bench(k_w2v_probe_pf<4, 6>, "V5 prefetch-6   ");
+
+  // NS-mode variants (pool indirection + in-kernel LCG, like k_w2v_ns)
+  long pool_n = 700000;
+  long* pool;
+  hipMalloc(&pool, pool_n * 8);
+  {
+    std::vector<long> hp(pool_n);
+    for (long i = 0; i < pool_n; ++i) hp[i] = zipf();
+    hipMemcpy(pool, hp.data(), pool_n * 8, hipMemcpyHostToDevice);
+  }
+  auto bench_ns = [&](auto kern, const char* name) {
+    kern<<<grid, BLOCK>>>(in_emb, out_emb, h_in.data() ? in_idx : in_idx,
+                          in_off, in_idx /*centers = in ids*/, pool, pool_n,
+                          neg, 12345ull, 0.025f, G, dim);
+    hipDeviceSynchronize();
+    hipEvent_t a, b;
+    hipEventCreate(&a); hipEventCreate(&b);
+    hipEventRecord(a);
+    for (int r = 0; r < 3; ++r)
+      kern<<<grid, BLOCK>>>(in_emb, out_emb, in_idx, in_off, in_idx, pool,
+                            pool_n, neg, 12345ull, 0.025f, G, dim);
+    hipEventRecord(b);
+    hipEventSynchronize(b);
+    float ms;
+    hipEventElapsedTime(&ms, a, b);
+    printf("%s: %.3f ms/launch (G=%d)\n", name, ms / 3, G);
+  };
+  bench_ns(k_w2v_probe_ns<4, false>, "V6 ns-pool      ");
+  bench_ns(k_w2v_probe_ns<4, true>, "V7 ns-pool-nin1 ");
   return 0;
 }
