@@ -75,3 +75,40 @@ def test_c_host_roundtrip(capi_so, tmp_path_factory):
                        timeout=300)
     assert r.returncode == 0, (r.returncode, r.stdout, r.stderr)
     assert "CAPI_OK" in r.stdout
+
+
+def test_c_host_roundtrip_sanitized(tmp_path_factory):
+    """the reference ships no sanitizer jobs; here the whole
+    native host path (C host + libmultiverso_amd) runs under
+    ASan+UBSan. Leaks are not checked (the embedded CPython interpreter
+    frees at process exit); memory errors and UB abort the run."""
+    import shutil
+    import sysconfig
+    import pybind11
+    tmp = tmp_path_factory.mktemp("capi_asan")
+    inc = os.path.join(REPO, "multiverso_amd", "capi")
+    so = tmp / "libmultiverso_amd.so"
+    py_inc = sysconfig.get_paths()["include"]
+    libdir = sysconfig.get_config_var("LIBDIR") or "/usr/lib"
+    pyver = f"python{sysconfig.get_python_version()}"
+    san = ["-fsanitize=address,undefined", "-fno-sanitize-recover=all",
+           "-fno-sanitize=vptr"]
+    subprocess.run(
+        ["g++", "-O1", "-g", "-std=c++17", "-shared", "-fPIC",
+         os.path.join(inc, "c_api.cpp"), f"-I{py_inc}",
+         f"-I{pybind11.get_include()}", f"-L{libdir}", f"-l{pyver}",
+         "-o", str(so)] + san, check=True)
+    csrc = tmp / "host.c"
+    csrc.write_text(C_HOST)
+    exe = tmp / "host"
+    subprocess.run(
+        ["gcc", "-O0", "-g", str(csrc), f"-I{inc}", f"-L{tmp}",
+         "-lmultiverso_amd", "-o", str(exe)] + san, check=True)
+    env = dict(os.environ)
+    env["LD_LIBRARY_PATH"] = f"{tmp}:" + env.get("LD_LIBRARY_PATH", "")
+    env["PYTHONPATH"] = REPO + ":" + env.get("PYTHONPATH", "")
+    env["ASAN_OPTIONS"] = "detect_leaks=0:abort_on_error=1"
+    r = subprocess.run([str(exe)], env=env, capture_output=True, text=True,
+                       timeout=600)
+    assert r.returncode == 0, (r.returncode, r.stdout[-500:], r.stderr[-2000:])
+    assert "CAPI_OK" in r.stdout
