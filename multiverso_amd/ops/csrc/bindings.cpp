@@ -18,6 +18,15 @@ void mv_launch_adagrad(float*, float*, const float*, float, float, float,
                        int64_t, hipStream_t);
 void mv_launch_sgd_copy(float*, const float*, float*, float, int64_t,
                         hipStream_t);
+void mv_launch_momentum_copy(float*, float*, const float*, float*, float,
+                             int64_t, hipStream_t);
+void mv_launch_adagrad_copy(float*, float*, const float*, float*, float,
+                            float, float, int64_t, hipStream_t);
+void mv_launch_dcasgd_copy(float*, float*, const float*, float*, float,
+                           float, int64_t, hipStream_t);
+void mv_launch_dcasgda_copy(float*, float*, float*, const float*, float*,
+                            float, float, float, float, int64_t,
+                            hipStream_t);
 void mv_launch_dcasgd(float*, float*, const float*, float, float, int64_t,
                       hipStream_t);
 void mv_launch_dcasgda(float*, float*, float*, const float*, float, float,
@@ -99,6 +108,57 @@ void sgd_copy_update(torch::Tensor data, torch::Tensor delta,
   mv_launch_sgd_copy(data.data_ptr<float>(), delta.data_ptr<float>(),
                      out.data_ptr<float>(), (float)sign, data.numel(),
                      cur_stream());
+}
+
+void momentum_copy_update(torch::Tensor data, torch::Tensor m,
+                          torch::Tensor delta, torch::Tensor out, double mu) {
+  check_f32(data, "data"); check_f32(m, "m"); check_f32(delta, "delta");
+  check_f32(out, "out");
+  TORCH_CHECK(data.numel() == delta.numel() && data.numel() == m.numel() &&
+              data.numel() == out.numel(), "size mismatch");
+  mv_launch_momentum_copy(data.data_ptr<float>(), m.data_ptr<float>(),
+                          delta.data_ptr<float>(), out.data_ptr<float>(),
+                          (float)mu, data.numel(), cur_stream());
+}
+
+void adagrad_copy_update(torch::Tensor data, torch::Tensor gsq,
+                         torch::Tensor delta, torch::Tensor out,
+                         double lr, double rho, double eps) {
+  check_f32(data, "data"); check_f32(gsq, "gsq"); check_f32(delta, "delta");
+  check_f32(out, "out");
+  TORCH_CHECK(data.numel() == delta.numel() && data.numel() == gsq.numel() &&
+              data.numel() == out.numel(), "size mismatch");
+  mv_launch_adagrad_copy(data.data_ptr<float>(), gsq.data_ptr<float>(),
+                         delta.data_ptr<float>(), out.data_ptr<float>(),
+                         (float)lr, (float)rho, (float)eps, data.numel(),
+                         cur_stream());
+}
+
+void dcasgd_copy_update(torch::Tensor data, torch::Tensor bak,
+                        torch::Tensor delta, torch::Tensor out,
+                        double lr, double lambda) {
+  check_f32(data, "data"); check_f32(bak, "bak"); check_f32(delta, "delta");
+  check_f32(out, "out");
+  TORCH_CHECK(data.numel() == delta.numel() && data.numel() == bak.numel() &&
+              data.numel() == out.numel(), "size mismatch");
+  mv_launch_dcasgd_copy(data.data_ptr<float>(), bak.data_ptr<float>(),
+                        delta.data_ptr<float>(), out.data_ptr<float>(),
+                        (float)lr, (float)lambda, data.numel(), cur_stream());
+}
+
+void dcasgda_copy_update(torch::Tensor data, torch::Tensor bak,
+                         torch::Tensor msq, torch::Tensor delta,
+                         torch::Tensor out, double lr, double lambda,
+                         double rho, double eps) {
+  check_f32(data, "data"); check_f32(bak, "bak"); check_f32(msq, "msq");
+  check_f32(delta, "delta"); check_f32(out, "out");
+  TORCH_CHECK(data.numel() == delta.numel() && data.numel() == bak.numel() &&
+              data.numel() == msq.numel() && data.numel() == out.numel(),
+              "size mismatch");
+  mv_launch_dcasgda_copy(data.data_ptr<float>(), bak.data_ptr<float>(),
+                         msq.data_ptr<float>(), delta.data_ptr<float>(),
+                         out.data_ptr<float>(), (float)lr, (float)lambda,
+                         (float)rho, (float)eps, data.numel(), cur_stream());
 }
 
 void dcasgd_update(torch::Tensor data, torch::Tensor bak, torch::Tensor delta,
@@ -258,6 +318,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_update", &sgd_update, "K2: data -= delta");
   m.def("momentum_update", &momentum_update, "K3: fused momentum update");
   m.def("adagrad_update", &adagrad_update, "K4: fused adagrad update");
+  m.def("momentum_copy_update", &momentum_copy_update,
+        "fused momentum Add+Get");
+  m.def("adagrad_copy_update", &adagrad_copy_update,
+        "fused adagrad Add+Get");
+  m.def("dcasgd_copy_update", &dcasgd_copy_update,
+        "fused dcasgd Add+Get");
+  m.def("dcasgda_copy_update", &dcasgda_copy_update,
+        "fused dcasgda Add+Get");
   m.def("sgd_copy_update", &sgd_copy_update,
         "fused Add+Get: data (+/-)= delta; out = data (saves the Get's "
         "shard re-read at N=1)");

@@ -168,6 +168,75 @@ __global__ void k_sgd_copy_tail(float* __restrict__ data,
   }
 }
 
+// Fused stateful-updater + Get variants: same saving as k_sgd_copy for
+// momentum/adagrad/dcasgd(a) — the updated weights stream to the Get
+// buffer in the same pass instead of a separate shard re-read.
+__global__ void k_momentum_copy_f4(v4f* __restrict__ data, v4f* __restrict__ m,
+                                   const v4f* __restrict__ delta,
+                                   v4f* __restrict__ out, float mu, int64_t n4) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float om = 1.0f - mu;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    v4f mm = mu * ntload(&m[i]) + om * ntload(&delta[i]);
+    ntstore(&m[i], mm);
+    v4f d = ntload(&data[i]) - mm;
+    ntstore(&data[i], d);
+    ntstore(&out[i], d);
+  }
+}
+
+__global__ void k_momentum_copy_tail(float* __restrict__ data,
+                                     float* __restrict__ m,
+                                     const float* __restrict__ delta,
+                                     float* __restrict__ out, float mu,
+                                     int64_t start, int64_t n) {
+  int64_t i = start + blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    float mm = mu * m[i] + (1.0f - mu) * delta[i];
+    m[i] = mm;
+    float d = data[i] - mm;
+    data[i] = d;
+    out[i] = d;
+  }
+}
+
+__global__ void k_adagrad_copy_f4(v4f* __restrict__ data, v4f* __restrict__ gsq,
+                                  const v4f* __restrict__ delta,
+                                  v4f* __restrict__ out,
+                                  float inv_lr, float rho, float eps,
+                                  int64_t n4) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    v4f d = ntload(&data[i]), G = ntload(&gsq[i]);
+    v4f g = ntload(&delta[i]) * inv_lr;
+    G += g * g;
+    d.x -= rho * g.x * __frsqrt_rn(G.x + eps);
+    d.y -= rho * g.y * __frsqrt_rn(G.y + eps);
+    d.z -= rho * g.z * __frsqrt_rn(G.z + eps);
+    d.w -= rho * g.w * __frsqrt_rn(G.w + eps);
+    ntstore(&gsq[i], G);
+    ntstore(&data[i], d);
+    ntstore(&out[i], d);
+  }
+}
+
+__global__ void k_adagrad_copy_tail(float* __restrict__ data,
+                                    float* __restrict__ gsq,
+                                    const float* __restrict__ delta,
+                                    float* __restrict__ out,
+                                    float inv_lr, float rho, float eps,
+                                    int64_t start, int64_t n) {
+  int64_t i = start + blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    float g = delta[i] * inv_lr;
+    float G = gsq[i] + g * g;
+    gsq[i] = G;
+    float d = data[i] - rho * g * __frsqrt_rn(G + eps);
+    data[i] = d;
+    out[i] = d;
+  }
+}
+
 // DC-ASGD updaters ("dcasgd"/"dcasgda", selected at reference
 // updater.cpp:51-54 from a submodule ABSENT from the snapshot
 // (.gitmodules:1-3, empty dir) — math reconstructed from the DC-ASGD
@@ -241,6 +310,82 @@ __global__ void k_dcasgda_tail(float* __restrict__ data, float* __restrict__ bak
         - lr * (g + lambda * __frsqrt_rn(m + eps) * g * g * (data[i] - bak[i]));
     data[i] = w;
     bak[i] = w;
+  }
+}
+
+__global__ void k_dcasgd_copy_f4(v4f* __restrict__ data, v4f* __restrict__ bak,
+                                 const v4f* __restrict__ delta,
+                                 v4f* __restrict__ out,
+                                 float lr, float lambda, int64_t n4) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    v4f w = ntload(&data[i]), b = ntload(&bak[i]);
+    v4f g = ntload(&delta[i]);
+    w -= lr * (g + lambda * g * g * (w - b));
+    ntstore(&data[i], w);
+    ntstore(&bak[i], w);
+    ntstore(&out[i], w);
+  }
+}
+
+__global__ void k_dcasgd_copy_tail(float* __restrict__ data,
+                                   float* __restrict__ bak,
+                                   const float* __restrict__ delta,
+                                   float* __restrict__ out,
+                                   float lr, float lambda,
+                                   int64_t start, int64_t n) {
+  int64_t i = start + blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    float g = delta[i];
+    float w = data[i] - lr * (g + lambda * g * g * (data[i] - bak[i]));
+    data[i] = w;
+    bak[i] = w;
+    out[i] = w;
+  }
+}
+
+__global__ void k_dcasgda_copy_f4(v4f* __restrict__ data, v4f* __restrict__ bak,
+                                  v4f* __restrict__ msq,
+                                  const v4f* __restrict__ delta,
+                                  v4f* __restrict__ out,
+                                  float lr, float lambda, float rho, float eps,
+                                  int64_t n4) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float om = 1.0f - rho;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    v4f w = ntload(&data[i]), b = ntload(&bak[i]);
+    v4f g = ntload(&delta[i]);
+    v4f m = rho * ntload(&msq[i]) + om * g * g;
+    ntstore(&msq[i], m);
+    v4f lam;
+    lam.x = lambda * __frsqrt_rn(m.x + eps);
+    lam.y = lambda * __frsqrt_rn(m.y + eps);
+    lam.z = lambda * __frsqrt_rn(m.z + eps);
+    lam.w = lambda * __frsqrt_rn(m.w + eps);
+    w -= lr * (g + lam * g * g * (w - b));
+    ntstore(&data[i], w);
+    ntstore(&bak[i], w);
+    ntstore(&out[i], w);
+  }
+}
+
+__global__ void k_dcasgda_copy_tail(float* __restrict__ data,
+                                    float* __restrict__ bak,
+                                    float* __restrict__ msq,
+                                    const float* __restrict__ delta,
+                                    float* __restrict__ out,
+                                    float lr, float lambda, float rho,
+                                    float eps, int64_t start, int64_t n) {
+  int64_t i = start + blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    float g = delta[i];
+    float m = rho * msq[i] + (1.0f - rho) * g * g;
+    msq[i] = m;
+    float w = data[i]
+        - lr * (g + lambda * __frsqrt_rn(m + eps) * g * g * (data[i] - bak[i]));
+    data[i] = w;
+    bak[i] = w;
+    out[i] = w;
   }
 }
 
@@ -367,6 +512,54 @@ void mv_launch_sgd_copy(float* data, const float* delta, float* out,
   int64_t tail = n - n4 * 4;
   if (tail) k_sgd_copy_tail<<<1, 64, 0, s>>>(data, delta, out, sign,
                                              n4 * 4, n);
+}
+
+void mv_launch_momentum_copy(float* data, float* m, const float* delta,
+                             float* out, float mu, int64_t n, hipStream_t s) {
+  int64_t n4 = n / 4;
+  if (n4) k_momentum_copy_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
+      (v4f*)data, (v4f*)m, (const v4f*)delta, (v4f*)out, mu, n4);
+  int64_t tail = n - n4 * 4;
+  if (tail) k_momentum_copy_tail<<<1, 64, 0, s>>>(data, m, delta, out, mu,
+                                                  n4 * 4, n);
+}
+
+void mv_launch_adagrad_copy(float* data, float* gsq, const float* delta,
+                            float* out, float lr, float rho, float eps,
+                            int64_t n, hipStream_t s) {
+  float inv_lr = 1.0f / lr;
+  int64_t n4 = n / 4;
+  if (n4) k_adagrad_copy_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
+      (v4f*)data, (v4f*)gsq, (const v4f*)delta, (v4f*)out, inv_lr, rho, eps,
+      n4);
+  int64_t tail = n - n4 * 4;
+  if (tail) k_adagrad_copy_tail<<<1, 64, 0, s>>>(data, gsq, delta, out,
+                                                 inv_lr, rho, eps, n4 * 4, n);
+}
+
+void mv_launch_dcasgd_copy(float* data, float* bak, const float* delta,
+                           float* out, float lr, float lambda, int64_t n,
+                           hipStream_t s) {
+  int64_t n4 = n / 4;
+  if (n4) k_dcasgd_copy_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
+      (v4f*)data, (v4f*)bak, (const v4f*)delta, (v4f*)out, lr, lambda, n4);
+  int64_t tail = n - n4 * 4;
+  if (tail) k_dcasgd_copy_tail<<<1, 64, 0, s>>>(data, bak, delta, out, lr,
+                                                lambda, n4 * 4, n);
+}
+
+void mv_launch_dcasgda_copy(float* data, float* bak, float* msq,
+                            const float* delta, float* out, float lr,
+                            float lambda, float rho, float eps, int64_t n,
+                            hipStream_t s) {
+  int64_t n4 = n / 4;
+  if (n4) k_dcasgda_copy_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
+      (v4f*)data, (v4f*)bak, (v4f*)msq, (const v4f*)delta, (v4f*)out, lr,
+      lambda, rho, eps, n4);
+  int64_t tail = n - n4 * 4;
+  if (tail) k_dcasgda_copy_tail<<<1, 64, 0, s>>>(data, bak, msq, delta, out,
+                                                 lr, lambda, rho, eps,
+                                                 n4 * 4, n);
 }
 
 void mv_launch_dcasgd(float* data, float* bak, const float* delta,

@@ -57,11 +57,12 @@ class MatrixTable(Table):
             self.shard.copy_(full[self.row_offset:
                                   self.row_offset + self.local_rows])
         self._make_updater(self.shard.view(-1))
-        # single-rank Add-deferral for stateless updaters: an Add
-        # immediately followed by a whole-table Get fuses into one kernel
-        # (k_sgd_copy) that writes both the shard and the Get buffer,
-        # saving the Get's shard re-read (0.5 GB/step on the headline
-        # config). Any other table op materializes the Add first (flush).
+        # single-rank Add-deferral: an Add immediately followed by a
+        # whole-table Get fuses into one updater kernel
+        # (updater.update_and_copy) that writes both the shard and the
+        # Get buffer, saving the Get's shard re-read (0.5 GB/step on the
+        # headline config). Any other table op materializes the Add
+        # first (flush).
         self._deferred = None  # (delta, option, delta._version)
 
     def _check_deferred(self, d) -> None:
@@ -91,11 +92,8 @@ class MatrixTable(Table):
             CHECK(out.numel() == self.num_row * self.num_col,
                   "Get buffer size mismatch")
             if out.is_contiguous():
-                from .. import ops
-                sign = -1.0 if self.updater_type == "sgd" else 1.0
                 with monitor("server.update"):
-                    ops.module(required=True).sgd_copy_update(
-                        self.shard.view(-1), d[0], out.view(-1), sign)
+                    self.updater.update_and_copy(d[0], d[1], out.view(-1))
                 return out
             # non-contiguous Get buffer: materialize, fall through
             with monitor("server.update"):
@@ -119,8 +117,7 @@ class MatrixTable(Table):
         CHECK(delta.numel() == self.num_row * self.num_col,
               "Add delta size mismatch")
         delta = delta.to(self.device, self.dtype).contiguous().view(-1)
-        if (self.zoo.size == 1 and self.shard.is_cuda
-                and self.updater_type in ("sgd", "default")):
+        if self.zoo.size == 1 and self.shard.is_cuda:
             self.flush()                 # at most one deferred Add
             self._deferred = (delta, option, delta._version)
             return Handle()

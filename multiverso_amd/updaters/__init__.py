@@ -80,6 +80,19 @@ class Updater:
         else:
             self.shard.add_(delta)
 
+    def update_and_copy(self, delta: torch.Tensor,
+                        option: Optional[AddOption],
+                        out: torch.Tensor) -> None:
+        """Fused Add+Get (single-rank fast path): apply the update AND
+        stream the updated shard into ``out`` in the same pass — saves
+        the Get's shard re-read. Every updater has a fused HIP kernel;
+        the CPU fallback composes update + copy."""
+        if self.shard.is_cuda:
+            _hip_ops().sgd_copy_update(self.shard, delta, out, 1.0)
+        else:
+            self.update(delta, option)
+            out.copy_(self.shard)
+
     def access(self, out: torch.Tensor) -> None:
         """K7: shard copy-out (updater.cpp:32-36)."""
         out.copy_(self.shard)
@@ -93,6 +106,13 @@ class SGDUpdater(Updater):
             _hip_ops().sgd_update(self.shard, delta)
         else:
             self.shard.sub_(delta)
+
+    def update_and_copy(self, delta, option, out) -> None:
+        if self.shard.is_cuda:
+            _hip_ops().sgd_copy_update(self.shard, delta, out, -1.0)
+        else:
+            self.update(delta, option)
+            out.copy_(self.shard)
 
 
 class MomentumUpdater(Updater):
@@ -110,6 +130,15 @@ class MomentumUpdater(Updater):
         else:
             self.smooth_gradient.mul_(mu).add_(delta, alpha=1.0 - mu)
             self.shard.sub_(self.smooth_gradient)
+
+    def update_and_copy(self, delta, option, out) -> None:
+        if self.shard.is_cuda:
+            mu = option.momentum if option else 0.0
+            _hip_ops().momentum_copy_update(self.shard, self.smooth_gradient,
+                                            delta, out, float(mu))
+        else:
+            self.update(delta, option)
+            out.copy_(self.shard)
 
 
 class AdaGradUpdater(Updater):
@@ -130,6 +159,16 @@ class AdaGradUpdater(Updater):
             g = delta / lr
             self.g_sqr.add_(g * g)
             self.shard.sub_(rho * g / torch.sqrt(self.g_sqr + self.EPS))
+
+    def update_and_copy(self, delta, option, out) -> None:
+        if self.shard.is_cuda:
+            opt = option or AddOption()
+            _hip_ops().adagrad_copy_update(self.shard, self.g_sqr, delta,
+                                           out, float(opt.learning_rate),
+                                           float(opt.rho), self.EPS)
+        else:
+            self.update(delta, option)
+            out.copy_(self.shard)
 
 
 class DCASGDUpdater(Updater):
@@ -170,6 +209,17 @@ class DCASGDUpdater(Updater):
             self.shard.sub_(lr * (g + lam * g * g * (self.shard - bak)))
             bak.copy_(self.shard)
 
+    def update_and_copy(self, delta, option, out) -> None:
+        if self.shard.is_cuda:
+            opt = option or AddOption()
+            bak = self._backup(opt.worker_id)
+            _hip_ops().dcasgd_copy_update(self.shard, bak, delta, out,
+                                          float(opt.learning_rate),
+                                          float(opt.lambda_))
+        else:
+            self.update(delta, option)
+            out.copy_(self.shard)
+
 
 class DCASGDAUpdater(DCASGDUpdater):
     """DC-ASGD-a ("dcasgda", updater.cpp:52): adaptive lambda — a running
@@ -197,6 +247,19 @@ class DCASGDAUpdater(DCASGDUpdater):
             lam_t = lam / torch.sqrt(self.mean_sqr + self.EPS)
             self.shard.sub_(lr * (g + lam_t * g * g * (self.shard - bak)))
             bak.copy_(self.shard)
+
+    def update_and_copy(self, delta, option, out) -> None:
+        if self.shard.is_cuda:
+            opt = option or AddOption()
+            bak = self._backup(opt.worker_id)
+            _hip_ops().dcasgda_copy_update(self.shard, bak, self.mean_sqr,
+                                           delta, out,
+                                           float(opt.learning_rate),
+                                           float(opt.lambda_),
+                                           float(opt.rho), self.EPS)
+        else:
+            self.update(delta, option)
+            out.copy_(self.shard)
 
 
 _REGISTRY: Dict[str, Type[Updater]] = {

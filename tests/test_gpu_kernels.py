@@ -222,3 +222,29 @@ def test_aggregate_large_gpu():
     torch.cuda.synchronize()
     assert float(out[0]) == 1.0 and float(out[-1]) == 1.0
     mv.shutdown()
+
+
+@pytest.mark.parametrize("updater,extra", [
+    ("momentum", 1), ("adagrad", 1), ("dcasgd", 1), ("dcasgda", 2)])
+def test_fused_stateful_add_get(updater, extra):
+    """Every stateful updater's Add+Get fuses; result must equal the
+    explicit update followed by a Get on a twin table."""
+    import multiverso_amd as mv
+    from multiverso_amd.updaters import AddOption
+    mv.init()
+    opt = AddOption(momentum=0.8, learning_rate=0.1, rho=0.2, lambda_=0.3)
+    a = mv.MatrixTable(500, 16, updater_type=updater)
+    b = mv.MatrixTable(500, 16, updater_type=updater)
+    delta = torch.randn(500, 16, device="cuda:0") * 0.1
+    # twin b: materialize the add (add_rows flushes), then plain get
+    b.add(delta.clone(), option=opt)
+    b.flush()
+    want = b.get()
+    # a: deferred add -> fused update_and_copy
+    a.add(delta.clone(), option=opt)
+    got = a.get()
+    torch.cuda.synchronize()
+    assert torch.allclose(got, want, rtol=1e-5, atol=1e-6), \
+        (got - want).abs().max()
+    assert torch.allclose(a.shard, b.shard, rtol=1e-5, atol=1e-6)
+    mv.shutdown()
