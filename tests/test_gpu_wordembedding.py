@@ -161,3 +161,98 @@ def test_w2v_gpu_end_to_end_learns():
     assert float(pos) > 0.6 and float(pos) - float(wrong) > 0.3, \
         (float(pos), float(wrong))
     mv.shutdown()
+
+
+def test_w2v_ns_kernel_adagrad_vs_reference():
+    """NS fast path with adagrad state, single group + 1-element pool:
+    exact sequential semantics vs the torch reference."""
+    from multiverso_amd import ops
+    from multiverso_amd.apps.wordembedding.model import _w2v_train_torch
+    hip = ops.module(required=True)
+    dim, neg = 96, 3
+    torch.manual_seed(6)
+    in_buf = torch.randn(1, dim) * 0.1
+    out_buf = torch.randn(2, dim) * 0.1
+    in_gsq = torch.zeros(1, dim)
+    out_gsq = torch.zeros(2, dim)
+    in_idx = torch.zeros(1, dtype=torch.int64)
+    in_off = torch.tensor([0, 1], dtype=torch.int32)
+    centers = torch.zeros(1, dtype=torch.int64)
+    pool = torch.tensor([1], dtype=torch.int64)
+    lr = 0.05
+
+    ref = [in_buf.clone(), out_buf.clone(), in_gsq.clone(), out_gsq.clone()]
+    out_idx = torch.tensor([0] + [1] * neg, dtype=torch.int64)
+    lab = torch.tensor([1.0] + [0.0] * neg)
+    out_off = torch.tensor([0, 1 + neg], dtype=torch.int32)
+    _w2v_train_torch(ref[0], ref[1], ref[2], ref[3], in_idx, in_off,
+                     out_idx, lab, out_off, lr, True, lr)
+
+    d = lambda t: t.cuda()
+    g = [d(in_buf), d(out_buf), d(in_gsq), d(out_gsq)]
+    hip.w2v_train_ns(g[0], g[1], g[2], g[3], d(in_idx), None, d(centers),
+                     d(pool), neg, 777, lr, True, False)
+    torch.cuda.synchronize()
+    for got, want in zip(g, ref):
+        assert torch.allclose(got.cpu(), want, rtol=1e-3, atol=1e-5), \
+            (got.cpu() - want).abs().max()
+
+
+def test_w2v_ns_kernel_cbow_ragged_inputs():
+    """NS fast path with ragged in_off (CBOW: several inputs averaged per
+    group), single group, deterministic pool."""
+    from multiverso_amd import ops
+    from multiverso_amd.apps.wordembedding.model import _w2v_train_torch
+    hip = ops.module(required=True)
+    dim, neg, nin = 64, 2, 3
+    torch.manual_seed(7)
+    in_buf = torch.randn(nin, dim) * 0.1
+    out_buf = torch.randn(2, dim) * 0.1
+    in_idx = torch.arange(nin, dtype=torch.int64)
+    in_off = torch.tensor([0, nin], dtype=torch.int32)
+    centers = torch.zeros(1, dtype=torch.int64)
+    pool = torch.tensor([1], dtype=torch.int64)
+    lr = 0.05
+
+    ref_in, ref_out = in_buf.clone(), out_buf.clone()
+    out_idx = torch.tensor([0] + [1] * neg, dtype=torch.int64)
+    lab = torch.tensor([1.0] + [0.0] * neg)
+    out_off = torch.tensor([0, 1 + neg], dtype=torch.int32)
+    _w2v_train_torch(ref_in, ref_out, None, None, in_idx, in_off,
+                     out_idx, lab, out_off, lr, False, lr)
+
+    d = lambda t: t.cuda()
+    gin, gout = d(in_buf), d(out_buf)
+    hip.w2v_train_ns(gin, gout, gin, gout, d(in_idx), d(in_off), d(centers),
+                     d(pool), neg, 555, lr, False, False)
+    torch.cuda.synchronize()
+    assert torch.allclose(gin.cpu(), ref_in, rtol=1e-4, atol=1e-5)
+    assert torch.allclose(gout.cpu(), ref_out, rtol=1e-4, atol=1e-5)
+
+
+def test_w2v_cbow_gpu_end_to_end_learns():
+    """CBOW + NS fast path end to end on GPU."""
+    import multiverso_amd as mv
+    from multiverso_amd.apps.wordembedding.model import (WordEmbedding,
+                                                         WordEmbeddingOption)
+    mv.init()
+    torch.manual_seed(1)
+    opt = WordEmbeddingOption(embedding_size=64, window=1, negative_num=5,
+                              cbow=True, init_learning_rate=0.1,
+                              total_words=10_000_000, seed=5,
+                              max_groups_per_launch=64)
+    model = WordEmbedding(opt, [100] * 20)
+    words = torch.stack([torch.arange(0, 20, 2).repeat(100),
+                         torch.arange(1, 20, 2).repeat(100)],
+                        dim=1).view(-1).cuda()
+    sids = (torch.arange(words.numel()) // 10).cuda()
+    for _ in range(10):
+        model.train_block(words, sids)
+    inp = model.input_table.get()
+    out = model.output_table.get()
+    evens = torch.arange(0, 20, 2, device="cuda")
+    pos = torch.sigmoid((inp[evens] * out[evens + 1]).sum(1)).mean()
+    wrong = torch.sigmoid((inp[evens] * out[evens.roll(1) + 1]).sum(1)).mean()
+    torch.cuda.synchronize()
+    assert float(pos) - float(wrong) > 0.25, (float(pos), float(wrong))
+    mv.shutdown()
