@@ -320,6 +320,97 @@ __global__ void k_w2v_probe_ns(float* __restrict__ in_emb,
   }
 }
 
+// VARIANT 8: V7 + depth-1 software pipeline over the output rows: the
+// next row's load issues before the current row's dot/update, hiding one
+// row-fetch latency per output while keeping VGPRs low (8 waves/SIMD).
+template <int DPL>
+__global__ void k_w2v_probe_ns_pf(float* __restrict__ in_emb,
+                                  float* __restrict__ out_emb,
+                                  const long* __restrict__ in_idx,
+                                  const int* __restrict__ in_off,
+                                  const long* __restrict__ centers,
+                                  const long* __restrict__ pool, long pool_n,
+                                  int neg, unsigned long long seed,
+                                  float lr, int G, int dim) {
+  int wid = (int)((blockIdx.x * (long)blockDim.x + threadIdx.x) >> 6);
+  int lane = threadIdx.x & 63;
+  int nwaves = (int)((gridDim.x * (long)blockDim.x) >> 6);
+  for (int g = wid; g < G; g += nwaves) {
+    float h[DPL], err[DPL];
+#pragma unroll
+    for (int d = 0; d < DPL; ++d) { h[d] = 0.f; err[d] = 0.f; }
+    {
+      const float* row = in_emb + in_idx[g] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) h[d] += row[c];
+      }
+    }
+    long pos = centers[g];
+    unsigned long long next_random = seed + (unsigned long long)g * 25214903917ull + 11ull;
+    // node sequence is computable ahead of the loop
+    long node = pos;
+    float label = 1.f;
+    float* w = out_emb + node * dim;
+    float wv[DPL], nv[DPL];
+#pragma unroll
+    for (int d = 0; d < DPL; ++d) {
+      int c = lane + 64 * d;
+      wv[d] = (c < dim) ? w[c] : 0.f;
+    }
+    for (int o = 0; o <= neg; ++o) {
+      // issue next row's loads before using this row's values
+      long nnode = -1;
+      float nlabel = 0.f;
+      int no = o + 1;
+      while (no <= neg) {
+        next_random = next_random * 25214903917ull + 11ull;
+        long cand = pool[(long)((next_random >> 8) % (unsigned long long)pool_n)];
+        if (cand != pos) { nnode = cand; break; }
+        ++no;
+      }
+      float* nw = nullptr;
+      if (nnode >= 0) {
+        nw = out_emb + nnode * dim;
+#pragma unroll
+        for (int d = 0; d < DPL; ++d) {
+          int c = lane + 64 * d;
+          nv[d] = (c < dim) ? nw[c] : 0.f;
+        }
+      }
+      float f = 0.f;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) f += h[d] * wv[d];
+#pragma unroll
+      for (int s = 32; s; s >>= 1) f += __shfl_xor(f, s, 64);
+      f = 1.f / (1.f + expf(-f));
+      float e = label - f;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) {
+          err[d] += e * wv[d];
+          w[c] = wv[d] + e * lr * h[d];
+        }
+      }
+      if (nnode < 0) break;
+      node = nnode; label = nlabel; w = nw;
+      o = no - 1;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) wv[d] = nv[d];
+    }
+    {
+      float* row = in_emb + in_idx[g] * dim;
+#pragma unroll
+      for (int d = 0; d < DPL; ++d) {
+        int c = lane + 64 * d;
+        if (c < dim) row[c] += lr * err[d];
+      }
+    }
+  }
+}
+
 int main(int argc, char** argv) {
   int G = argc > 1 ? atoi(argv[1]) : 1 << 21;
   int V = argc > 2 ? atoi(argv[2]) : 1000000;  // vocab
@@ -421,5 +512,6 @@ int main(int argc, char** argv) {
   };
   bench_ns(k_w2v_probe_ns<4, false>, "V6 ns-pool      ");
   bench_ns(k_w2v_probe_ns<4, true>, "V7 ns-pool-nin1 ");
+  bench_ns(k_w2v_probe_ns_pf<4>, "V8 ns-pipelined ");
   return 0;
 }
