@@ -38,11 +38,43 @@ class ArrayTable(Table):
         off, cnt = self.spec.range_of(self.zoo.server_id)
         self.shard = torch.zeros(cnt, dtype=dtype, device=self.device)
         self._make_updater(self.shard)
+        # single-rank Add deferral + Add/Get fusion — same mechanism and
+        # semantics as MatrixTable (see matrix_table.py)
+        self._deferred = None  # (delta, option, delta._version)
+
+    def _check_deferred(self, d) -> None:
+        CHECK(d[0]._version == d[2],
+              "the delta tensor passed to Add was mutated in place before "
+              "the Add was applied (single-GPU deferred-Add fast path "
+              "snapshots at the next table op; pass a fresh tensor)")
+
+    def flush(self) -> None:
+        d = self._deferred
+        if d is not None:
+            self._deferred = None
+            self._check_deferred(d)
+            with monitor("server.update"):
+                self.updater.update(d[0], d[1])
+        super().flush()
 
     # ---- worker ops ----
     def get(self, out: Optional[torch.Tensor] = None,
             async_op: bool = False):
         """Whole-table Get (array_table.cpp:24-66 semantics)."""
+        d = self._deferred
+        if d is not None and not async_op:
+            self._deferred = None
+            self._check_deferred(d)
+            if out is None:
+                out = torch.empty(self.size, dtype=self.dtype,
+                                  device=self.device)
+            CHECK(out.numel() == self.size, "Get buffer size mismatch")
+            if out.is_contiguous():
+                with monitor("server.update"):
+                    self.updater.update_and_copy(d[0], d[1], out.view(-1))
+                return out
+            with monitor("server.update"):
+                self.updater.update(d[0], d[1])
         self.flush()
         if out is None:
             out = torch.empty(self.size, dtype=self.dtype, device=self.device)
@@ -61,6 +93,10 @@ class ArrayTable(Table):
         array_table.cpp:116-127)."""
         CHECK(delta.numel() == self.size, "Add delta size mismatch")
         delta = delta.to(self.device, self.dtype).contiguous().view(-1)
+        if self.zoo.size == 1 and self.shard.is_cuda:
+            self.flush()                 # at most one deferred Add
+            self._deferred = (delta, option, delta._version)
+            return Handle()
         with monitor("worker.add"):
             chunk, h = reduce_scatter_delta(delta, self.spec, 1,
                                             async_op=async_op)
@@ -88,6 +124,7 @@ class ArrayTable(Table):
         self.zoo.barrier()
 
     def load(self, path: str) -> None:
+        self.flush()
         import numpy as np
         arr = np.fromfile(path, dtype=str(self.dtype).replace("torch.", ""))
         CHECK(arr.size == self.size, "checkpoint size mismatch")

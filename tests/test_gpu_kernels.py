@@ -248,3 +248,26 @@ def test_fused_stateful_add_get(updater, extra):
         (got - want).abs().max()
     assert torch.allclose(a.shard, b.shard, rtol=1e-5, atol=1e-6)
     mv.shutdown()
+
+
+def test_array_fused_add_get():
+    import multiverso_amd as mv
+    mv.init()
+    t = mv.ArrayTable(10_000, updater_type="sgd")
+    delta = torch.randn(10_000, device="cuda:0")
+    t.add(delta)
+    got = t.get()
+    torch.cuda.synchronize()
+    assert torch.allclose(got, -delta, rtol=1e-6, atol=1e-7)
+    assert torch.allclose(t.shard, -delta, rtol=1e-6, atol=1e-7)
+    # checkpoint after a deferred add must flush first
+    t.add(delta)
+    import tempfile, os
+    p = os.path.join(tempfile.gettempdir(), "arr_fused.bin")
+    t.store(p)
+    t2 = mv.ArrayTable(10_000, updater_type="sgd")
+    t2.load(p)
+    torch.cuda.synchronize()
+    assert torch.allclose(t2.get(), -2 * delta, rtol=1e-5, atol=1e-6)
+    os.remove(p)
+    mv.shutdown()
