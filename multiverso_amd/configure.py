@@ -101,7 +101,6 @@ def _define_core_flags() -> None:
     define_flag("log_level", "info", "debug|info|error|fatal")
     # MI355X-native additions
     define_flag("bucket_mb", 64, "collective bucket size (MiB) for sharded Add/Get over xGMI")
-    define_flag("comm_stream", True, "run collectives on a side HIP stream")
     define_flag("deterministic", False, "force deterministic reduction order in updaters")
 
 

@@ -48,7 +48,6 @@ class Zoo:
         self.size = 1
         self.role = Role.ALL
         self.device: torch.device = torch.device("cpu")
-        self.comm_stream = None  # side HIP stream for collectives
         self._tables: Dict[int, object] = {}
         self._next_table_id = 0
         self._owns_pg = False
@@ -152,9 +151,6 @@ class Zoo:
             self.size = dist.get_world_size()
         else:
             self.rank, self.size = 0, 1
-
-        if cuda:
-            self.comm_stream = torch.cuda.Stream(device=self.device)
 
         self.started = True
         log.debug(f"Zoo started: rank {self.rank}/{self.size} "
