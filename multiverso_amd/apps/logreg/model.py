@@ -126,21 +126,25 @@ class PSModel:
         local = pulled.clone()
         total_loss = 0.0
         adagrad = self.table.updater_type == "adagrad"
-        for b in batches:
-            lidx = torch.searchsorted(union, b.keys)
-            w_rows = local[lidx]
-            grad, loss = self.objective.gradient(b, w_rows)
-            total_loss += loss
-            if self.is_ftrl:
-                ops.scatter_add_rows(local, lidx, grad, -1.0)
-            elif adagrad:
-                # server-side adagrad consumes lr-scaled deltas; locally
-                # approximate with plain sgd steps for within-chunk vis.
-                lr = self.sched.next_lr()
-                ops.scatter_add_rows(local, lidx, grad, -lr)
-            else:
-                lr = self.sched.next_lr()
-                ops.scatter_add_rows(local, lidx, grad, -lr)
+        if self._fused_sigmoid(local):
+            total_loss = self._train_chunk_fused(batches, union, local)
+        else:
+            for b in batches:
+                lidx = torch.searchsorted(union, b.keys)
+                w_rows = local[lidx]
+                grad, loss = self.objective.gradient(b, w_rows)
+                total_loss += loss
+                if self.is_ftrl:
+                    ops.scatter_add_rows(local, lidx, grad, -1.0)
+                elif adagrad:
+                    # server-side adagrad consumes lr-scaled deltas;
+                    # locally approximate with plain sgd steps for
+                    # within-chunk vis.
+                    lr = self.sched.next_lr()
+                    ops.scatter_add_rows(local, lidx, grad, -lr)
+                else:
+                    lr = self.sched.next_lr()
+                    ops.scatter_add_rows(local, lidx, grad, -lr)
         if adagrad:
             delta = pulled - local   # = sum(lr*grad); server g=delta/lr
             opt = mv.AddOption(learning_rate=1.0, rho=self.cfg.learning_rate)
@@ -149,6 +153,44 @@ class PSModel:
             # server updater 'sgd': w -= delta; push accumulated movement
             self.table.add_rows(union, pulled - local)
         return total_loss / max(len(batches), 1)
+
+    def _fused_sigmoid(self, local: torch.Tensor) -> bool:
+        """Fused K13/K14 minibatch kernels: sigmoid objective, single
+        output column, GPU. Other objectives use the torch path."""
+        from .objective import SigmoidObjective
+        return (local.is_cuda and self.cols == 1
+                and type(self.objective) is SigmoidObjective)
+
+    def _train_chunk_fused(self, batches, union, local) -> float:
+        """Two HIP kernels per minibatch (k_lr_sigmoid_fwd/_scatter)
+        instead of ~20 torch ops; numerics match objective.gradient +
+        scatter (tests/test_gpu_kernels.py). Returns summed per-batch
+        mean losses (one host sync per chunk instead of per minibatch)."""
+        from multiverso_amd import ops as _ops
+        hip = _ops.module(required=True)
+        from .objective import L1Regular, L2Regular
+        reg = self.objective.regular
+        reg_type, reg_coef = 0, 0.0
+        if isinstance(reg, L1Regular):
+            reg_type, reg_coef = 1, reg.coef
+        elif isinstance(reg, L2Regular):
+            reg_type, reg_coef = 2, reg.coef
+        wflat = local.view(-1)
+        loss_acc = torch.zeros((), device=self.device)
+        for b in batches:
+            lidx = torch.searchsorted(union, b.keys)
+            ptr32 = b.ptr.to(torch.int32)
+            B = b.size
+            err = torch.empty(B, device=self.device)
+            lossb = torch.empty(B, device=self.device)
+            wts = None if b.weights is None else b.weights.float()
+            hip.lr_sigmoid_forward(wflat, lidx, b.vals, ptr32,
+                                   b.labels.float(), wts, err, lossb)
+            lr = self.sched.next_lr()
+            hip.lr_sigmoid_scatter(wflat, lidx, b.vals, ptr32, err, lr,
+                                   reg_type, reg_coef)
+            loss_acc += lossb.mean()
+        return float(loss_acc)
 
     def predict(self, batch: Batch) -> torch.Tensor:
         batch = batch.to(self.device)

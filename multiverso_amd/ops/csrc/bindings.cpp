@@ -42,6 +42,12 @@ void mv_launch_w2v_ns(float*, float*, float*, float*, const int64_t*,
                       const int*, const int64_t*, const int64_t*, int64_t,
                       int, uint64_t, float, int64_t, int64_t, int, int,
                       hipStream_t);
+void mv_launch_lr_sigmoid_fwd(const float*, const int64_t*, const float*,
+                              const int*, const float*, const float*,
+                              float*, float*, int64_t, hipStream_t);
+void mv_launch_lr_sigmoid_scatter(float*, const int64_t*, const float*,
+                                  const int*, const float*, float, int,
+                                  float, int64_t, hipStream_t);
 void mv_launch_row_scatter_adagrad(float*, float*, const float*,
                                    const int64_t*, float, float, float,
                                    int64_t, int64_t, hipStream_t);
@@ -310,6 +316,48 @@ void w2v_train_ns(torch::Tensor in_emb, torch::Tensor out_emb,
                    cur_stream());
 }
 
+void lr_sigmoid_forward(torch::Tensor w, torch::Tensor keys,
+                        torch::Tensor vals, torch::Tensor ptr,
+                        torch::Tensor labels,
+                        c10::optional<torch::Tensor> wts,
+                        torch::Tensor err, torch::Tensor loss) {
+  check_f32(w, "w"); check_f32(vals, "vals"); check_f32(labels, "labels");
+  check_f32(err, "err"); check_f32(loss, "loss");
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64, "keys must be int64");
+  TORCH_CHECK(ptr.scalar_type() == torch::kInt32, "ptr must be int32");
+  int64_t B = ptr.numel() - 1;
+  TORCH_CHECK(labels.numel() == B && err.numel() == B && loss.numel() == B,
+              "batch size mismatch");
+  TORCH_CHECK(keys.numel() == vals.numel(), "keys/vals mismatch");
+  const float* wp = nullptr;
+  if (wts.has_value()) {
+    check_f32(*wts, "wts");
+    TORCH_CHECK(wts->numel() == B, "weights size mismatch");
+    wp = wts->data_ptr<float>();
+  }
+  mv_launch_lr_sigmoid_fwd(w.data_ptr<float>(), keys.data_ptr<int64_t>(),
+                           vals.data_ptr<float>(), ptr.data_ptr<int>(),
+                           labels.data_ptr<float>(), wp,
+                           err.data_ptr<float>(), loss.data_ptr<float>(), B,
+                           cur_stream());
+}
+
+void lr_sigmoid_scatter(torch::Tensor w, torch::Tensor keys,
+                        torch::Tensor vals, torch::Tensor ptr,
+                        torch::Tensor err, double lr, int64_t reg_type,
+                        double reg_coef) {
+  check_f32(w, "w"); check_f32(vals, "vals"); check_f32(err, "err");
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64, "keys must be int64");
+  TORCH_CHECK(ptr.scalar_type() == torch::kInt32, "ptr must be int32");
+  int64_t B = ptr.numel() - 1;
+  TORCH_CHECK(err.numel() == B, "err size mismatch");
+  mv_launch_lr_sigmoid_scatter(w.data_ptr<float>(), keys.data_ptr<int64_t>(),
+                               vals.data_ptr<float>(), ptr.data_ptr<int>(),
+                               err.data_ptr<float>(), (float)lr,
+                               (int)reg_type, (float)reg_coef, B,
+                               cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -337,6 +385,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_gather_out", &row_gather_out, "K6 (preallocated out)");
   m.def("row_scatter_add", &row_scatter_add,
         "K5: shard[rows[i]] += sign*vals[i] (atomic)");
+  m.def("lr_sigmoid_forward", &lr_sigmoid_forward,
+        "K13 fused: per-sample CSR dot + sigmoid + error/loss");
+  m.def("lr_sigmoid_scatter", &lr_sigmoid_scatter,
+        "K14 fused: w[keys] -= lr*(vals*err + reg), atomic");
   m.def("row_scatter_adagrad", &row_scatter_adagrad,
         "K15: keyed adagrad update on owned shard rows");
   m.def("w2v_train", &w2v_train,
