@@ -148,10 +148,11 @@ class PSModel:
         if adagrad:
             delta = pulled - local   # = sum(lr*grad); server g=delta/lr
             opt = mv.AddOption(learning_rate=1.0, rho=self.cfg.learning_rate)
-            self.table.add_rows(union, delta, option=opt)
+            self.table.add_rows(union, delta, option=opt,
+                                assume_unique=True)
         else:
             # server updater 'sgd': w -= delta; push accumulated movement
-            self.table.add_rows(union, pulled - local)
+            self.table.add_rows(union, pulled - local, assume_unique=True)
         return total_loss / max(len(batches), 1)
 
     def _fused_sigmoid(self, local: torch.Tensor) -> bool:

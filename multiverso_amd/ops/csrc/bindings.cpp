@@ -34,7 +34,7 @@ void mv_launch_dcasgda(float*, float*, float*, const float*, float, float,
 void mv_launch_row_gather(float*, const float*, const int64_t*, int64_t,
                           int64_t, hipStream_t);
 void mv_launch_row_scatter_add(float*, const float*, const int64_t*, float,
-                               int64_t, int64_t, hipStream_t);
+                               int64_t, int64_t, int, hipStream_t);
 void mv_launch_w2v(float*, float*, float*, float*, const int64_t*, const int*,
                    const int64_t*, const float*, const int*, float, int64_t,
                    int64_t, int, int, hipStream_t);
@@ -50,7 +50,7 @@ void mv_launch_lr_sigmoid_scatter(float*, const int64_t*, const float*,
                                   float, int64_t, hipStream_t);
 void mv_launch_row_scatter_adagrad(float*, float*, const float*,
                                    const int64_t*, float, float, float,
-                                   int64_t, int64_t, hipStream_t);
+                                   int64_t, int64_t, int, hipStream_t);
 }
 
 namespace {
@@ -214,7 +214,8 @@ void row_gather_out(torch::Tensor out, torch::Tensor shard, torch::Tensor rows) 
 }
 
 void row_scatter_add(torch::Tensor shard, torch::Tensor rows,
-                     torch::Tensor vals, double sign) {
+                     torch::Tensor vals, double sign,
+                     bool assume_unique = false) {
   check_f32(shard, "shard"); check_f32(vals, "vals");
   TORCH_CHECK(shard.dim() == 2, "shard must be 2-D");
   TORCH_CHECK(rows.scalar_type() == torch::kInt64 && rows.is_cuda() &&
@@ -222,12 +223,14 @@ void row_scatter_add(torch::Tensor shard, torch::Tensor rows,
   TORCH_CHECK(vals.numel() == rows.numel() * shard.size(1), "vals size mismatch");
   mv_launch_row_scatter_add(shard.data_ptr<float>(), vals.data_ptr<float>(),
                             rows.data_ptr<int64_t>(), (float)sign,
-                            rows.numel(), shard.size(1), cur_stream());
+                            rows.numel(), shard.size(1),
+                            assume_unique ? 1 : 0, cur_stream());
 }
 
 void row_scatter_adagrad(torch::Tensor shard, torch::Tensor gsq,
                          torch::Tensor rows, torch::Tensor vals,
-                         double lr, double rho, double eps) {
+                         double lr, double rho, double eps,
+                         bool assume_unique = false) {
   check_f32(shard, "shard"); check_f32(gsq, "gsq"); check_f32(vals, "vals");
   TORCH_CHECK(shard.dim() == 2, "shard must be 2-D");
   TORCH_CHECK(gsq.sizes() == shard.sizes(), "gsq shape mismatch");
@@ -237,7 +240,7 @@ void row_scatter_adagrad(torch::Tensor shard, torch::Tensor gsq,
   mv_launch_row_scatter_adagrad(
       shard.data_ptr<float>(), gsq.data_ptr<float>(), vals.data_ptr<float>(),
       rows.data_ptr<int64_t>(), (float)lr, (float)rho, (float)eps,
-      rows.numel(), shard.size(1), cur_stream());
+      rows.numel(), shard.size(1), assume_unique ? 1 : 0, cur_stream());
 }
 
 void w2v_train(torch::Tensor in_emb, torch::Tensor out_emb,
@@ -384,13 +387,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_gather", &row_gather, "K6: out[i] = shard[rows[i]]");
   m.def("row_gather_out", &row_gather_out, "K6 (preallocated out)");
   m.def("row_scatter_add", &row_scatter_add,
-        "K5: shard[rows[i]] += sign*vals[i] (atomic)");
+        "K5: shard[rows[i]] += sign*vals[i] (atomic; assume_unique=True "
+        "uses plain stores)", py::arg("shard"), py::arg("rows"),
+        py::arg("vals"), py::arg("sign"), py::arg("assume_unique") = false);
   m.def("lr_sigmoid_forward", &lr_sigmoid_forward,
         "K13 fused: per-sample CSR dot + sigmoid + error/loss");
   m.def("lr_sigmoid_scatter", &lr_sigmoid_scatter,
         "K14 fused: w[keys] -= lr*(vals*err + reg), atomic");
   m.def("row_scatter_adagrad", &row_scatter_adagrad,
-        "K15: keyed adagrad update on owned shard rows");
+        "K15: keyed adagrad update on owned shard rows "
+        "(assume_unique=True uses plain stores)",
+        py::arg("shard"), py::arg("gsq"), py::arg("rows"), py::arg("vals"),
+        py::arg("lr"), py::arg("rho"), py::arg("eps"),
+        py::arg("assume_unique") = false);
   m.def("w2v_train", &w2v_train,
         "K9-K11: fused word2vec block training (skip-gram/CBOW, NS/HS, "
         "optional adagrad)");

@@ -449,6 +449,23 @@ __global__ void k_row_scatter_add(float* __restrict__ shard,
   }
 }
 
+// unique-rows variants: when the caller guarantees no duplicate row ids
+// (e.g. a sorted-unique union pushed from ONE rank) the atomics are
+// unnecessary — plain read-modify-write is substantially faster for
+// scattered few-column rows.
+__global__ void k_row_scatter_add_u(float* __restrict__ shard,
+                                    const float* __restrict__ vals,
+                                    const int64_t* __restrict__ rows,
+                                    float sign, int64_t nrows, int64_t cols) {
+  int64_t total = nrows * cols;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
+    int64_t r = i / cols, c = i % cols;
+    int64_t k = rows[r] * cols + c;
+    shard[k] += sign * vals[i];
+  }
+}
+
 // ---------------------------------------------------------------------------
 // extern "C" launchers
 // ---------------------------------------------------------------------------
@@ -599,10 +616,15 @@ void mv_launch_row_gather(float* out, const float* shard, const int64_t* rows,
 
 void mv_launch_row_scatter_add(float* shard, const float* vals,
                                const int64_t* rows, float sign,
-                               int64_t nrows, int64_t cols, hipStream_t s) {
+                               int64_t nrows, int64_t cols,
+                               int assume_unique, hipStream_t s) {
   if (!nrows || !cols) return;
-  k_row_scatter_add<<<grid_for(nrows * cols), BLOCK, 0, s>>>(
-      shard, vals, rows, sign, nrows, cols);
+  if (assume_unique)
+    k_row_scatter_add_u<<<grid_for(nrows * cols), BLOCK, 0, s>>>(
+        shard, vals, rows, sign, nrows, cols);
+  else
+    k_row_scatter_add<<<grid_for(nrows * cols), BLOCK, 0, s>>>(
+        shard, vals, rows, sign, nrows, cols);
 }
 
 }  // extern "C"
@@ -1018,11 +1040,35 @@ __global__ void k_row_scatter_adagrad(float* __restrict__ shard,
   }
 }
 
+__global__ void k_row_scatter_adagrad_u(float* __restrict__ shard,
+                                        float* __restrict__ gsq,
+                                        const float* __restrict__ vals,
+                                        const int64_t* __restrict__ rows,
+                                        float inv_lr, float rho, float eps,
+                                        int64_t nrows, int64_t cols) {
+  int64_t total = nrows * cols;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
+    int64_t r = i / cols, c = i % cols;
+    int64_t k = rows[r] * cols + c;
+    float g = vals[i] * inv_lr;
+    if (g != 0.0f) {
+      float G = gsq[k] + g * g;
+      gsq[k] = G;
+      shard[k] -= rho * g * __frsqrt_rn(G + eps);
+    }
+  }
+}
+
 extern "C" void mv_launch_row_scatter_adagrad(
     float* shard, float* gsq, const float* vals, const int64_t* rows,
     float lr, float rho, float eps, int64_t nrows, int64_t cols,
-    hipStream_t s) {
+    int assume_unique, hipStream_t s) {
   if (!nrows || !cols) return;
-  k_row_scatter_adagrad<<<grid_for(nrows * cols), BLOCK, 0, s>>>(
-      shard, gsq, vals, rows, 1.0f / lr, rho, eps, nrows, cols);
+  if (assume_unique)
+    k_row_scatter_adagrad_u<<<grid_for(nrows * cols), BLOCK, 0, s>>>(
+        shard, gsq, vals, rows, 1.0f / lr, rho, eps, nrows, cols);
+  else
+    k_row_scatter_adagrad<<<grid_for(nrows * cols), BLOCK, 0, s>>>(
+        shard, gsq, vals, rows, 1.0f / lr, rho, eps, nrows, cols);
 }

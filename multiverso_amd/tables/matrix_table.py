@@ -149,7 +149,8 @@ class MatrixTable(Table):
 
     def _scatter_update_local(self, local_ids: torch.Tensor,
                               vals: torch.Tensor,
-                              option: Optional[AddOption]) -> None:
+                              option: Optional[AddOption],
+                              assume_unique: bool = False) -> None:
         """K5 on the owned shard. default adds, sgd subtracts (the per-row
         updater dispatch of matrix_table.cpp:406-412); adagrad applies the
         keyed K15 form (duplicate rows: GPU races benignly via atomics,
@@ -163,7 +164,8 @@ class MatrixTable(Table):
                 from .. import ops
                 ops.module(required=True).row_scatter_adagrad(
                     self.shard, gsq, local_ids, vals2.contiguous(),
-                    opt.learning_rate, opt.rho, self.updater.EPS)
+                    opt.learning_rate, opt.rho, self.updater.EPS,
+                    assume_unique)
             else:
                 urows, inv = torch.unique(local_ids, return_inverse=True)
                 agg = torch.zeros(urows.numel(), self.num_col,
@@ -182,7 +184,8 @@ class MatrixTable(Table):
         if self.shard.is_cuda:
             from .. import ops
             ops.module(required=True).row_scatter_add(
-                self.shard, local_ids, vals.contiguous(), sign)
+                self.shard, local_ids, vals.contiguous(), sign,
+                assume_unique)
         else:
             self.shard.index_add_(0, local_ids,
                                   vals.view(-1, self.num_col) * sign)
@@ -218,8 +221,13 @@ class MatrixTable(Table):
         return out
 
     def add_rows(self, row_ids, values: torch.Tensor,
-                 option: Optional[AddOption] = None) -> None:
-        """Row-subset Add (matrix_table.cpp:265-309 partition path)."""
+                 option: Optional[AddOption] = None,
+                 assume_unique: bool = False) -> None:
+        """Row-subset Add (matrix_table.cpp:265-309 partition path).
+        ``assume_unique=True`` asserts the caller's row_ids have no
+        duplicates (e.g. a sorted-unique union) — at world size 1 the
+        scatter then skips atomics; with multiple ranks incoming ids can
+        still collide across ranks, so atomics stay."""
         self.flush()   # a deferred whole-table Add must land first
         ids = torch.as_tensor(row_ids, dtype=torch.int64, device=self.device)
         vals = values.to(self.device, self.dtype).contiguous()
@@ -231,8 +239,9 @@ class MatrixTable(Table):
             if in_ids.numel():
                 local = self._local_rows_of(in_ids)
                 with monitor("server.update_rows"):
-                    self._scatter_update_local(local, in_vals.view(
-                        -1, self.num_col), option)
+                    self._scatter_update_local(
+                        local, in_vals.view(-1, self.num_col), option,
+                        assume_unique=assume_unique and self.zoo.size == 1)
 
     # ---- checkpoint (matrix_table.cpp:457-464) ----
     def store(self, path: str) -> None:
