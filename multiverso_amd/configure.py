@@ -102,6 +102,8 @@ def _define_core_flags() -> None:
     # MI355X-native additions
     define_flag("bucket_mb", 64, "collective bucket size (MiB) for sharded Add/Get over xGMI")
     define_flag("deterministic", False, "force deterministic reduction order in updaters")
+    define_flag("sparse_filter", True, "SparseFilter on stale-row reply payloads "
+                "(sparse_matrix_table.cpp:148-153 parity)")
 
 
 _define_core_flags()

@@ -35,6 +35,12 @@ class SparseMatrixTable(MatrixTable):
                  updater_type: Optional[str] = None,
                  random_init=None) -> None:
         super().__init__(num_row, num_col, dtype, updater_type, random_init)
+        # SparseFilter on the stale-row reply payloads (the reference
+        # filters every outgoing sparse-table bundle,
+        # sparse_matrix_table.cpp:148-153): per-destination segments are
+        # compressed when >50% zero; byte accounting on the Dashboard.
+        from ..configure import get_flag
+        self.use_sparse_filter = bool(get_flag("sparse_filter"))
         nw = self.zoo.num_workers
         # up_to_date[w, r] == True -> worker w has the current row r
         self.up_to_date = torch.zeros(nw, self.local_rows, dtype=torch.bool,
@@ -74,6 +80,45 @@ class SparseMatrixTable(MatrixTable):
                 # is never-stale-safe and costs one extra row re-pull)
                 self.up_to_date[:, local] = False
 
+    def _filtered_value_exchange(self, served: torch.Tensor,
+                                 send_rows, recv_rows) -> torch.Tensor:
+        """Value half of the stale-row exchange with SparseFilter applied
+        per destination segment (FilterIn on send, FilterOut on receive —
+        quantization_util.h:95-154 semantics over the collective)."""
+        from .. import sparse_filter as sf
+        device = self.device
+        packed_parts, packed_sizes = [], []
+        off = 0
+        one = torch.ones(1, dtype=torch.float32, device=device)
+        zero = torch.zeros(1, dtype=torch.float32, device=device)
+        for n in send_rows:
+            seg = served[off:off + n].reshape(-1)
+            off += n
+            payload, compressed = sf.filter_in(seg)
+            # self-describing wire: 1-float header = compressed flag
+            payload = torch.cat([one if compressed else zero,
+                                 payload.float()])
+            packed_parts.append(payload)
+            packed_sizes.append(payload.numel())
+        send_buf = (torch.cat(packed_parts) if packed_parts else
+                    torch.empty(0, device=device))
+        scnt = torch.tensor(packed_sizes, dtype=torch.int64, device=device)
+        rcnt = torch.empty_like(scnt)
+        dist.all_to_all_single(rcnt, scnt)
+        rsizes = rcnt.tolist()
+        recv_buf = torch.empty(sum(rsizes), dtype=torch.float32,
+                               device=device)
+        dist.all_to_all_single(recv_buf, send_buf, rsizes, packed_sizes)
+        outs, off = [], 0
+        for n, sz in zip(recv_rows, rsizes):
+            payload = recv_buf[off:off + sz]
+            off += sz
+            compressed = bool(payload[0] > 0.5)
+            outs.append(sf.filter_out(payload[1:], compressed,
+                                      n * self.num_col))
+        return (torch.cat(outs) if outs else
+                torch.empty(0, dtype=self.dtype, device=device))
+
     # ---- stale-filtered whole-table get ----
     def get_into(self, cache: torch.Tensor) -> int:
         """Overwrite only the rows of ``cache`` that are stale for this
@@ -109,8 +154,12 @@ class SparseMatrixTable(MatrixTable):
                       if all_ids.numel() else
                       torch.empty(0, self.num_col, dtype=self.dtype,
                                   device=self.device))
-            got_vals = all_to_all_values(served.view(-1), send_rows,
-                                         recv_rows, self.num_col)
+            if self.use_sparse_filter:
+                got_vals = self._filtered_value_exchange(
+                    served, send_rows, recv_rows)
+            else:
+                got_vals = all_to_all_values(served.view(-1), send_rows,
+                                             recv_rows, self.num_col)
             # 3) write into cache; mark fresh on the server side
             if got_ids.numel():
                 cache[got_ids] = got_vals.view(-1, self.num_col)

@@ -150,3 +150,33 @@ def _checkpoint_dist(rank, world):
 
 def test_checkpoint_dist():
     run_dist(_checkpoint_dist, 2)
+
+
+def _sparse_filter_wire(rank, world):
+    """Stale-row replies travel SparseFilter-compressed (the initial pull
+    of an all-zeros table is ~free) and decompress exactly."""
+    import multiverso_amd as mv
+    import torch
+    mv.init(sync=True)
+    from multiverso_amd.dashboard import Dashboard
+    t = mv.SparseMatrixTable(10, 4)
+    assert t.use_sparse_filter
+    cache = torch.full((10, 4), -1.0)
+    n = t.get_into(cache)   # all-zero table: payloads compress to ~nothing
+    assert n == 10 and torch.equal(cache, torch.zeros(10, 4))
+    if rank == 0:
+        t.add_rows([2], torch.tensor([[1.0, 0.0, 0.0, 2.0]]))
+    else:
+        t.add_rows([], torch.zeros(0, 4))
+    n = t.get_into(cache)
+    assert n == 1, n
+    assert torch.equal(cache[2], torch.tensor([1.0, 0.0, 0.0, 2.0]))
+    bytes_in = Dashboard.get("sparse_filter.bytes_in").elapsed_ms
+    bytes_out = Dashboard.get("sparse_filter.bytes_out").elapsed_ms
+    assert bytes_in > 0 and bytes_out < bytes_in  # compression happened
+    mv.shutdown()
+
+
+def test_sparse_filter_wire_dist():
+    from conftest import run_dist
+    run_dist(_sparse_filter_wire, 2)
