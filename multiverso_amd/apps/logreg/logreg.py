@@ -37,7 +37,8 @@ class LogReg:
         last_loss = 0.0
         for epoch in range(cfg.train_epoch):
             it = batches if batches is not None else SampleReader(
-                cfg.train_file, cfg.minibatch_size, cfg.reader_type).batches()
+                cfg.train_file, cfg.minibatch_size, cfg.reader_type,
+                input_size=cfg.input_size).batches()
             t0 = time.perf_counter()
             nsamples = 0
             chunk: List[Batch] = []
@@ -71,7 +72,8 @@ class LogReg:
              output_file: Optional[str] = None):
         cfg = self.cfg
         it = batches if batches is not None else SampleReader(
-            cfg.test_file, cfg.minibatch_size, cfg.reader_type).batches()
+            cfg.test_file, cfg.minibatch_size, cfg.reader_type,
+            input_size=cfg.input_size).batches()
         correct = total = 0
         loss_sum = 0.0
         nb = 0

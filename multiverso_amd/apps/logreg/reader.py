@@ -20,60 +20,76 @@ import torch
 from .objective import Batch
 
 
-def parse_text_lines(lines: List[str], weighted: bool) -> Batch:
+def parse_text_lines(lines: List[str], weighted: bool,
+                     bias_key: int = -1) -> Batch:
+    """default: "label k:v k:v ..." (SampleReader::ParseLine,
+    reader.cpp:177-217); weight: "label weight k:v ..." with the weight
+    folded into the values (WeightedSampleReader, :243-261). Every sample
+    gets the bias feature key ``bias_key`` (= row_size-1) with value 1
+    (:195-196, :216), when bias_key >= 0."""
     keys: List[int] = []
     vals: List[float] = []
     ptr = [0]
     labels: List[float] = []
-    weights: List[float] = []
     for line in lines:
         parts = line.split()
         if not parts:
             continue
         labels.append(float(parts[0]))
         i = 1
+        w = 1.0
         if weighted:
-            weights.append(float(parts[1]))
+            w = float(parts[1])
             i = 2
         for kv in parts[i:]:
             k, _, v = kv.partition(":")
             keys.append(int(k))
-            vals.append(float(v) if v else 1.0)
+            vals.append((float(v) if v else 1.0) * w)
+        if bias_key >= 0:
+            keys.append(bias_key)
+            vals.append(1.0)
         ptr.append(len(keys))
     return Batch(torch.tensor(keys, dtype=torch.int64),
                  torch.tensor(vals, dtype=torch.float32),
                  torch.tensor(ptr, dtype=torch.int64),
-                 torch.tensor(labels, dtype=torch.float32),
-                 torch.tensor(weights, dtype=torch.float32)
-                 if weighted else None)
+                 torch.tensor(labels, dtype=torch.float32))
 
 
-def write_bsparse(path: str, samples: List[Tuple[int, List[Tuple[int, float]]]]) -> None:
-    """bsparse binary: per sample [int32 label][int32 nnz]
-    ([int64 key][float32 val])*nnz."""
+def write_bsparse(path: str,
+                  samples: List[Tuple[int, float, List[int]]]) -> None:
+    """Reference bsparse binary (BSparseSampleReader::ParseSample,
+    reader.cpp:391-427): per sample [size_t nnz][int32 label]
+    [double weight][size_t key]*nnz; every key's value IS the weight."""
     with open(path, "wb") as f:
-        for label, kvs in samples:
-            f.write(struct.pack("<ii", label, len(kvs)))
-            for k, v in kvs:
-                f.write(struct.pack("<qf", k, v))
+        for label, weight, ks in samples:
+            f.write(struct.pack("<qid", len(ks), label, weight))
+            for k in ks:
+                f.write(struct.pack("<q", k))
 
 
-def read_bsparse_batches(path: str, minibatch: int) -> Iterator[Batch]:
+def read_bsparse_batches(path: str, minibatch: int,
+                         bias_key: int = -1) -> Iterator[Batch]:
+    """Parses the reference layout; values = sample weight for every
+    key, plus the bias key (= row_size-1, value weight) when
+    bias_key >= 0 — reader.cpp:420-423."""
     keys: List[int] = []
     vals: List[float] = []
     ptr = [0]
     labels: List[float] = []
     with open(path, "rb") as f:
         while True:
-            head = f.read(8)
-            if len(head) < 8:
+            head = f.read(20)
+            if len(head) < 20:
                 break
-            label, nnz = struct.unpack("<ii", head)
+            nnz, label, weight = struct.unpack("<qid", head)
             labels.append(float(label))
             for _ in range(nnz):
-                k, v = struct.unpack("<qf", f.read(12))
+                (k,) = struct.unpack("<q", f.read(8))
                 keys.append(k)
-                vals.append(v)
+                vals.append(weight)
+            if bias_key >= 0:
+                keys.append(bias_key)
+                vals.append(weight)
             ptr.append(len(keys))
             if len(labels) >= minibatch:
                 yield Batch(torch.tensor(keys, dtype=torch.int64),
@@ -93,16 +109,20 @@ class SampleReader:
     minibatch Batches (the reference's parse thread, reader.cpp)."""
 
     def __init__(self, path: str, minibatch: int, reader_type: str = "default",
-                 buffer_batches: int = 8) -> None:
+                 buffer_batches: int = 8, input_size: int = 0) -> None:
         self.path = path
         self.minibatch = minibatch
         self.reader_type = reader_type
         self.buffer_batches = buffer_batches
+        # bias feature at row_size-1 (reader.cpp:196): requires knowing
+        # the model width; 0 disables (synthetic/benchmark data)
+        self.bias_key = input_size - 1 if input_size > 0 else -1
 
     def _produce(self, q: Queue) -> None:
         try:
             if self.reader_type == "bsparse":
-                for b in read_bsparse_batches(self.path, self.minibatch):
+                for b in read_bsparse_batches(self.path, self.minibatch,
+                                              self.bias_key):
                     q.put(b)
             else:
                 weighted = self.reader_type == "weight"
@@ -112,10 +132,11 @@ class SampleReader:
                         if line.strip():
                             lines.append(line)
                         if len(lines) >= self.minibatch:
-                            q.put(parse_text_lines(lines, weighted))
+                            q.put(parse_text_lines(lines, weighted,
+                                                   self.bias_key))
                             lines = []
                 if lines:
-                    q.put(parse_text_lines(lines, weighted))
+                    q.put(parse_text_lines(lines, weighted, self.bias_key))
         finally:
             q.put(None)
 

@@ -43,22 +43,38 @@ def test_text_reader(tmp_path):
     assert b.size == 2
     assert b.keys.tolist() == [0, 3, 2]
     assert b.labels.tolist() == [1.0, 0.0]
+    # with input_size the bias feature (row_size-1, value 1) is appended
+    # to every sample (reference reader.cpp:195-196,216)
+    b2 = list(SampleReader(str(p), 2, input_size=100).batches())[0]
+    assert b2.keys.tolist() == [0, 3, 99, 2, 99]
+    assert b2.vals.tolist() == [0.5, 1.5, 1.0, 1.0, 1.0]
 
 
 def test_weighted_reader(tmp_path):
+    # WeightedSampleReader folds the weight into the values
+    # (reader.cpp:261: value * weight)
     p = tmp_path / "w.txt"
-    p.write_text("1 2.0 0:1\n")
-    b = list(SampleReader(str(p), 4, "weight").batches())[0]
-    assert b.weights.tolist() == [2.0]
+    p.write_text("1 2.0 0:1 3:0.5\n")
+    b = list(SampleReader(str(p), 4, "weight", input_size=10).batches())[0]
+    assert b.keys.tolist() == [0, 3, 9]
+    assert b.vals.tolist() == [2.0, 1.0, 1.0]
 
 
 def test_bsparse_roundtrip(tmp_path):
+    # reference layout: [size_t nnz][int32 label][double weight]
+    # [size_t keys...]; value = weight everywhere incl. bias
     p = str(tmp_path / "b.bin")
-    write_bsparse(p, [(1, [(5, 0.5), (9, 1.0)]), (0, [(2, 2.0)])])
-    batches = list(read_bsparse_batches(p, 10))
+    write_bsparse(p, [(1, 0.5, [5, 9]), (0, 2.0, [2])])
+    batches = list(read_bsparse_batches(p, 10, bias_key=99))
     assert batches[0].size == 2
-    assert batches[0].keys.tolist() == [5, 9, 2]
-    assert batches[0].vals.tolist() == [0.5, 1.0, 2.0]
+    assert batches[0].keys.tolist() == [5, 9, 99, 2, 99]
+    assert batches[0].vals.tolist() == [0.5, 0.5, 0.5, 2.0, 2.0]
+    # byte-layout check against the reference struct sizes
+    import struct, os
+    raw = open(p, "rb").read()
+    assert len(raw) == (20 + 16) + (20 + 8)
+    nnz, label, weight = struct.unpack_from("<qid", raw, 0)
+    assert (nnz, label, weight) == (2, 1, 0.5)
 
 
 def _train_local(cfg, n_batches=60):
