@@ -69,11 +69,11 @@ class Dictionary:
                 f.write(f"{w} {c}\n")
 
     @classmethod
-    def load(cls, path: str) -> "Dictionary":
+    def load(cls, path: str, min_count: int = 0) -> "Dictionary":
         d = cls()
         with open(path) as f:
             for line in f:
                 parts = line.split()
-                if len(parts) == 2:
+                if len(parts) == 2 and int(parts[1]) >= min_count:
                     d.insert(parts[0], int(parts[1]))
         return d

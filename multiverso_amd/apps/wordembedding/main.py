@@ -62,7 +62,10 @@ def main(argv=None) -> None:
     if args.stopwords and args.sw_file:
         stop = set(open(args.sw_file).read().split())
     if args.read_vocab:
-        dictionary = Dictionary.load(args.read_vocab)
+        # LoadVocab applies min_count after loading
+        # (distributed_wordembedding.cpp:442)
+        dictionary = Dictionary.load(args.read_vocab,
+                                     min_count=args.min_count)
     else:
         dictionary = Dictionary.build(tokenize_file(args.train_file),
                                       min_count=args.min_count,
