@@ -119,22 +119,27 @@ class SampleReader:
         self.bias_key = input_size - 1 if input_size > 0 else -1
 
     def _produce(self, q: Queue) -> None:
+        # train_file may name several files separated by ';'
+        # (reference configure.h:55)
+        paths = [p for p in self.path.split(";") if p]
         try:
             if self.reader_type == "bsparse":
-                for b in read_bsparse_batches(self.path, self.minibatch,
-                                              self.bias_key):
-                    q.put(b)
+                for path in paths:
+                    for b in read_bsparse_batches(path, self.minibatch,
+                                                  self.bias_key):
+                        q.put(b)
             else:
                 weighted = self.reader_type == "weight"
                 lines: List[str] = []
-                with open(self.path) as f:
-                    for line in f:
-                        if line.strip():
-                            lines.append(line)
-                        if len(lines) >= self.minibatch:
-                            q.put(parse_text_lines(lines, weighted,
-                                                   self.bias_key))
-                            lines = []
+                for path in paths:
+                    with open(path) as f:
+                        for line in f:
+                            if line.strip():
+                                lines.append(line)
+                            if len(lines) >= self.minibatch:
+                                q.put(parse_text_lines(lines, weighted,
+                                                       self.bias_key))
+                                lines = []
                 if lines:
                     q.put(parse_text_lines(lines, weighted, self.bias_key))
         finally:
