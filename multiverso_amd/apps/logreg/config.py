@@ -24,7 +24,13 @@ class LogRegConfig:
     updater_type: str = "sgd"           # default | sgd | ftrl | adagrad
     sparse: bool = False
     use_ps: bool = False
-    pipeline: bool = False
+    # reference default true (configure.h:84): its effect — overlapping
+    # the next chunk's model pull and data parse with training — is
+    # structural here: the SampleReader parses ahead on its own thread
+    # and single-rank pulls are stream-ordered behind the training
+    # kernels (no host sync), so the flag is accepted for config parity
+    # and changes nothing.
+    pipeline: bool = True
     sync_frequency: int = 1
     reader_type: str = "default"        # default | weight | bsparse
     train_file: str = ""
