@@ -261,3 +261,43 @@ def test_w2v_cbow_gpu_end_to_end_learns():
     assert float(pos) > 0.75 and float(pos) - float(wrong) > 0.15, \
         (float(pos), float(wrong))
     mv.shutdown()
+
+
+def test_w2v_hs_gpu_end_to_end_learns():
+    """Hierarchical-softmax path (ragged kernel, Huffman labels) end to
+    end on GPU: after training, a (center, context) pair's Huffman-path
+    probability must beat a mismatched pair's."""
+    import multiverso_amd as mv
+    from multiverso_amd.apps.wordembedding.model import (WordEmbedding,
+                                                         WordEmbeddingOption)
+    mv.init()
+    torch.manual_seed(2)
+    opt = WordEmbeddingOption(embedding_size=64, window=1, negative_num=0,
+                              hs=True, init_learning_rate=0.1,
+                              total_words=10_000_000, seed=7,
+                              max_groups_per_launch=64)
+    model = WordEmbedding(opt, [100] * 20)
+    words = torch.stack([torch.arange(0, 20, 2).repeat(100),
+                         torch.arange(1, 20, 2).repeat(100)],
+                        dim=1).view(-1).cuda()
+    sids = (torch.arange(words.numel()) // 10).cuda()
+    for _ in range(10):
+        model.train_block(words, sids)
+    inp = model.input_table.get()
+    out = model.output_table.get()
+
+    def path_logprob(ctx, center):
+        info = model.huffman.labels[center]
+        h = inp[ctx]
+        lp = 0.0
+        for node, code in zip(info.point, info.code):
+            p = torch.sigmoid((h * out[node]).sum())
+            # label = 1 - code (model.py _build_hs_tensors)
+            lp += float(torch.log((1 - code) * p + code * (1 - p) + 1e-9))
+        return lp
+
+    good = sum(path_logprob(e, e + 1) for e in range(0, 20, 2)) / 10
+    bad = sum(path_logprob(e, ((e + 3) % 20) | 1) for e in range(0, 20, 2)) / 10
+    torch.cuda.synchronize()
+    assert good > bad + 0.5, (good, bad)
+    mv.shutdown()
