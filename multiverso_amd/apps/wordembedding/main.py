@@ -107,11 +107,28 @@ def main(argv=None) -> None:
 
     t0 = time.perf_counter()
     trained = 0
+    empty = (torch.empty(0, dtype=torch.int64),
+             torch.empty(0, dtype=torch.int64))
     for epoch in range(args.epoch):
         reader = TextBlockReader(args.train_file, dictionary,
                                  args.data_block_size, mv.rank(), mv.size())
-        for words, sids in block_stream(reader):
-            trained += model.train_block(words, sids)
+        stream = block_stream(reader)
+        while True:
+            blk = next(stream, None)
+            if mv.size() > 1:
+                # ranks may hold different block counts (uneven corpus
+                # stripes); train_block is collective, so every rank
+                # keeps calling — with an empty block once exhausted —
+                # until ALL ranks are done.
+                have = torch.tensor([0.0 if blk is None else 1.0])
+                mv.aggregate(have)
+                if float(have[0]) == 0.0:
+                    break
+                if blk is None:
+                    blk = empty
+            elif blk is None:
+                break
+            trained += model.train_block(*blk)
             model.sync_word_count()
             if mv.rank() == 0:
                 dt = time.perf_counter() - t0

@@ -290,16 +290,21 @@ class WordEmbedding:
         # host-side randint/gather/label/ragged build at all.
         ns_fast = (opt.negative_num > 0 and not opt.hs
                    and self.device.type == "cuda")
+        empty_block = False
         if ns_fast:
             in_idx, in_off, centers = self.build_pairs(words, sent_ids)
-            if centers.numel() == 0:
-                return int(words.numel())
+            empty_block = centers.numel() == 0
             out_idx = out_label = out_off = None
         else:
             in_idx, in_off, out_idx, out_label, out_off = \
                 self.build_groups(words, sent_ids)
-            if in_idx.numel() == 0 or out_idx.numel() == 0:
-                return int(words.numel())
+            empty_block = in_idx.numel() == 0 or out_idx.numel() == 0
+        if empty_block and mv.size() == 1:
+            return int(words.numel())
+        # multi-rank: an empty block still participates in the collective
+        # pull/push below with zero-sized exchanges — ranks can hold
+        # different block counts (uneven corpus stripes) as long as the
+        # driver loop keeps calling collectively (main.py).
 
         if mv.size() == 1:
             # Single-rank fast path: every row is local, and
@@ -357,11 +362,12 @@ class WordEmbedding:
             gbufs = (igq, igq.clone(), ogq, ogq.clone())
 
         if ns_fast:
-            self._train_kernel_ns(in_buf, out_buf, gbufs[0], gbufs[2],
-                                  in_local,
-                                  in_off if opt.cbow else None,
-                                  out_local, pool_local)
-        else:
+            if out_local.numel() or in_local.numel():
+                self._train_kernel_ns(in_buf, out_buf, gbufs[0], gbufs[2],
+                                      in_local,
+                                      in_off if opt.cbow else None,
+                                      out_local, pool_local)
+        elif out_local.numel():
             self._train_kernel(in_buf, out_buf, gbufs[0], gbufs[2],
                                in_local, in_off, out_local,
                                out_label.float(), out_off)
