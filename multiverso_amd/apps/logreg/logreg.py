@@ -41,25 +41,51 @@ class LogReg:
                 input_size=cfg.input_size).batches()
             t0 = time.perf_counter()
             nsamples = 0
-            chunk: List[Batch] = []
             losses: List[float] = []
-            for b in it:
-                nsamples += b.size
-                if cfg.use_ps:
-                    chunk.append(b)
-                    if len(chunk) >= cfg.sync_frequency:
-                        losses.append(self.model.train_chunk(chunk))
-                        chunk = []
-                else:
-                    losses.append(self.model.update(b))
+
+            def show(b_size: int) -> None:
                 if (cfg.show_time_per_sample
-                        and nsamples % cfg.show_time_per_sample < b.size):
+                        and nsamples % cfg.show_time_per_sample < b_size):
                     dt = time.perf_counter() - t0
                     mv.log.info(f"[logreg] epoch {epoch} samples {nsamples} "
                                 f"loss {sum(losses)/max(len(losses),1):.5f} "
                                 f"{dt/max(nsamples,1)*1e6:.2f} us/sample")
-            if chunk:
-                losses.append(self.model.train_chunk(chunk))
+
+            if not cfg.use_ps:
+                for b in it:
+                    nsamples += b.size
+                    losses.append(self.model.update(b))
+                    show(b.size)
+            else:
+                def chunks():
+                    chunk: List[Batch] = []
+                    for b in it:
+                        chunk.append(b)
+                        if len(chunk) >= cfg.sync_frequency:
+                            yield chunk
+                            chunk = []
+                    if chunk:
+                        yield chunk
+
+                ci = chunks()
+                while True:
+                    c = next(ci, None)
+                    if mv.size() > 1:
+                        # train_chunk is collective and ranks may hold
+                        # different sample counts (per-rank train files,
+                        # configure.h:55): exhausted ranks keep joining
+                        # with empty chunks until every rank is done.
+                        import torch
+                        have = torch.tensor([0.0 if c is None else 1.0])
+                        mv.aggregate(have)
+                        if float(have[0]) == 0.0:
+                            break
+                        c = c or []
+                    elif c is None:
+                        break
+                    nsamples += sum(b.size for b in c)
+                    losses.append(self.model.train_chunk(c))
+                    show(sum(b.size for b in c) or 1)
             last_loss = sum(losses) / max(len(losses), 1)
             mv.log.info(f"[logreg] epoch {epoch} done: avg loss "
                         f"{last_loss:.5f}")

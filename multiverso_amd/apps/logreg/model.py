@@ -121,7 +121,9 @@ class PSModel:
         the pulled snapshot (within-chunk updates visible locally), push
         the summed delta (ps_model.cpp:172-203)."""
         batches = [b.to(self.device) for b in batches]
-        union = torch.unique(torch.cat([b.keys for b in batches]))
+        union = (torch.unique(torch.cat([b.keys for b in batches]))
+                 if batches else
+                 torch.empty(0, dtype=torch.int64, device=self.device))
         pulled = self.table.get_rows(union)
         local = pulled.clone()
         total_loss = 0.0

@@ -191,3 +191,36 @@ def _ps_dist(rank, world):
 
 def test_ps_dist():
     run_dist(_ps_dist, 2)
+
+
+def _logreg_uneven_files(rank, world, tmpdir):
+    """Per-rank train files with different sample counts (the reference's
+    multi-file deployment): the collective chunk loop must not
+    desynchronize."""
+    import os
+    import torch
+    import multiverso_amd as mv
+    from multiverso_amd.apps.logreg.config import LogRegConfig
+    from multiverso_amd.apps.logreg.logreg import LogReg
+    mv.init(sync=True)
+    path = os.path.join(tmpdir, f"r{rank}.txt")
+    n = 40 if rank == 0 else 12
+    with open(path, "w") as f:
+        for i in range(n):
+            f.write(f"{i % 2} {i % 50}:1 {(i * 7) % 50}:0.5\n")
+    cfg = LogRegConfig(input_size=64, output_size=1,
+                       objective_type="sigmoid", updater_type="sgd",
+                       learning_rate=0.1, minibatch_size=8,
+                       train_epoch=1, reader_type="default", use_ps=True,
+                       sync_frequency=2, train_file=path, test_file=path)
+    lr = LogReg(cfg)
+    lr.train()
+    acc, _ = lr.test()
+    assert 0.0 <= acc <= 1.0
+    mv.shutdown()
+
+
+def test_logreg_uneven_files_dist(tmp_path):
+    import functools
+    from conftest import run_dist
+    run_dist(functools.partial(_logreg_uneven_files, tmpdir=str(tmp_path)), 2)
