@@ -104,8 +104,29 @@ class LogReg:
         loss_sum = 0.0
         nb = 0
         out = open(output_file, "w") if output_file else None
-        for b in it:
+        collective = cfg.use_ps and mv.size() > 1
+        itr = iter(it)
+        while True:
+            b = next(itr, None)
+            if collective:
+                # predict pulls rows collectively; ranks may hold
+                # different test-set sizes -- empty-batch participation
+                # until all ranks are done (same protocol as train)
+                import torch
+                have = torch.tensor([0.0 if b is None else 1.0])
+                mv.aggregate(have)
+                if float(have[0]) == 0.0:
+                    break
+                if b is None:
+                    e = torch.empty(0, dtype=torch.int64)
+                    b = Batch(e, torch.empty(0),
+                              torch.zeros(1, dtype=torch.int64),
+                              torch.empty(0))
+            elif b is None:
+                break
             p = self.model.predict(b)
+            if b.size == 0:
+                continue
             correct += self.model.objective.correct(b.to(p.device), p)
             loss_sum += self.model.objective.loss(b.to(p.device), p)
             nb += 1
