@@ -92,3 +92,23 @@ def _model_average(rank, world):
 
 def test_model_average():
     run_dist(_model_average, 2)
+
+
+def _bucketed_aggregate(rank, world):
+    import multiverso_amd as mv
+    import torch
+    mv.init()
+    t = torch.arange(1000, dtype=torch.float32)
+    out = mv.aggregate(t.clone(), bucket_mb=0)       # single call
+    assert torch.equal(out, t * world)
+    # force the bucket pipeline: ~4KB tensor with a tiny bucket is still
+    # one bucket at 1 MiB floor, so build >1 MiB of data
+    big = torch.ones(600_000)
+    out2 = mv.aggregate(big.clone(), bucket_mb=1)    # 2.4 MB -> 3 buckets
+    assert torch.equal(out2, big * world)
+    mv.shutdown()
+
+
+def test_bucketed_aggregate_dist():
+    from conftest import run_dist
+    run_dist(_bucketed_aggregate, 2)

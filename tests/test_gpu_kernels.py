@@ -322,3 +322,26 @@ def test_lr_fused_kernels_vs_torch():
     hip.lr_sigmoid_scatter(w2, k2, v2, p2, e2, lr, 2, 0.01)
     torch.cuda.synchronize()
     assert torch.allclose(w2, ref2, rtol=1e-5, atol=1e-6)
+
+
+def test_get_noncontiguous_buffer_fallback():
+    """Deferred Add + Get into a non-contiguous buffer must materialize
+    the add and fall through to the gather path (fusion needs a
+    contiguous out)."""
+    import multiverso_amd as mv
+    mv.init()
+    t = mv.MatrixTable(64, 16, updater_type="sgd")
+    delta = torch.ones(64, 16, device="cuda:0")
+    t.add(delta)
+    big = torch.zeros(64, 32, device="cuda:0")
+    view = big[:, :16]          # non-contiguous
+    out = t.get(out=view)
+    torch.cuda.synchronize()
+    assert torch.allclose(out, -delta)
+    # deferred add + ASYNC get: deferral materializes via flush
+    t.add(delta)
+    out2, h = t.get(async_op=True)
+    h.wait()
+    torch.cuda.synchronize()
+    assert torch.allclose(out2, -2 * delta)
+    mv.shutdown()

@@ -186,3 +186,42 @@ def test_dashboard(env):
         pass
     assert mv.Dashboard.get("unit.test").count == 1
     assert "unit.test" in mv.Dashboard.display()
+
+
+def test_table_smaller_than_world_checks(env):
+    from multiverso_amd.log import FatalError
+    try:
+        mv.ArrayTable(0)
+        assert False, "expected CHECK failure"
+    except FatalError:
+        pass
+
+
+def test_get_with_user_buffer(env):
+    t = mv.ArrayTable(16, updater_type="default")
+    t.add(torch.arange(16, dtype=torch.float32))
+    buf = torch.empty(16)
+    out = t.get(out=buf)
+    assert out is buf
+    assert torch.equal(buf, torch.arange(16, dtype=torch.float32))
+
+
+def test_handler_numpy_inputs(env):
+    h = mv.ArrayTableHandler(8)
+    h.add(np.ones(8))
+    got = h.get()
+    assert torch.equal(got.cpu(), torch.ones(8))
+    m = mv.MatrixTableHandler(3, 4)
+    m.add(np.full((3, 4), 2.0))
+    assert torch.equal(m.get().cpu(), torch.full((3, 4), 2.0))
+    m.add(np.ones((2, 4)), row_ids=[0, 2])
+    assert torch.equal(m.get(row_ids=[0]).cpu(), torch.full((1, 4), 3.0))
+
+
+def test_updater_default_options(env):
+    # option=None must behave like a default AddOption for every updater
+    for u in ("momentum", "adagrad", "dcasgd", "dcasgda"):
+        t = mv.ArrayTable(4, updater_type=u)
+        t.add(torch.ones(4))
+        got = t.get()
+        assert torch.isfinite(got).all(), u
