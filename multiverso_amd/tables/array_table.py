@@ -76,12 +76,20 @@ class ArrayTable(Table):
             with monitor("server.update"):
                 self.updater.update(d[0], d[1])
         self.flush()
+        user_out = out
         if out is None:
             out = torch.empty(self.size, dtype=self.dtype, device=self.device)
         CHECK(out.numel() == self.size, "Get buffer size mismatch")
+        if not out.is_contiguous():
+            out = torch.empty(self.size, dtype=self.dtype, device=self.device)
         with monitor("worker.get"):
             h = allgather_shards(out.view(-1), self.shard, self.spec, 1,
                                  async_op=async_op)
+        if out is not user_out and user_out is not None:
+            if async_op:
+                h.wait()
+            user_out.copy_(out.view_as(user_out))
+            out = user_out
         if async_op:
             self._track(h)
             return out, h

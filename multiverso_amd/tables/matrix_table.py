@@ -99,14 +99,25 @@ class MatrixTable(Table):
             with monitor("server.update"):
                 self.updater.update(d[0], d[1])
         self.flush()
+        user_out = out
         if out is None:
             out = torch.empty(self.num_row, self.num_col, dtype=self.dtype,
                               device=self.device)
         CHECK(out.numel() == self.num_row * self.num_col,
               "Get buffer size mismatch")
+        if not out.is_contiguous():
+            # gather needs a flat view; stage and copy into the user's
+            # strided buffer afterwards
+            out = torch.empty(self.num_row, self.num_col, dtype=self.dtype,
+                              device=self.device)
         with monitor("worker.get"):
             h = allgather_shards(out.view(-1), self.shard.view(-1), self.spec,
                                  self.num_col, async_op=async_op)
+        if out is not user_out and user_out is not None:
+            if async_op:
+                h.wait()
+            user_out.copy_(out.view_as(user_out))
+            out = user_out
         if async_op:
             self._track(h)
             return out, h
