@@ -94,14 +94,14 @@ def _define_core_flags() -> None:
     define_flag("sync", False, "BSP synchronous server (vector-clocked)")
     define_flag("backup_worker_ratio", 0.0, "vestigial in reference; kept for parity")
     define_flag("updater_type", "default", "default|sgd|momentum|adagrad")
-    define_flag("omp_threads", 4, "CPU-fallback update threads (reference updater.cpp:18)")
+    define_flag("omp_threads", 0, "CPU-fallback intra-op threads (reference updater.cpp:18; 0 = torch default)")
     define_flag("allocator_type", "smart", "kept for parity; caching allocator is torch's")
     define_flag("allocator_alignment", 16, "kept for parity")
     define_flag("logtostderr", False, "log to stderr instead of stdout")
     define_flag("log_level", "info", "debug|info|error|fatal")
     # MI355X-native additions
     define_flag("bucket_mb", 64, "collective bucket size (MiB) for sharded Add/Get over xGMI")
-    define_flag("deterministic", False, "force deterministic reduction order in updaters")
+    define_flag("deterministic", False, "pre-aggregate duplicate rows before keyed scatter (fixed reduction order instead of atomics)")
     define_flag("sparse_filter", True, "SparseFilter on stale-row reply payloads "
                 "(sparse_matrix_table.cpp:148-153 parity)")
 

@@ -166,6 +166,20 @@ class MatrixTable(Table):
         updater dispatch of matrix_table.cpp:406-412); adagrad applies the
         keyed K15 form (duplicate rows: GPU races benignly via atomics,
         CPU pre-aggregates duplicates — both keep the G accumulate)."""
+        from ..configure import get_flag
+        if (get_flag("deterministic") and not assume_unique
+                and local_ids.numel()):
+            # fixed reduction order: pre-sum duplicate rows (sorted
+            # segments) and scatter without atomics
+            order = torch.argsort(local_ids, stable=True)
+            sids = local_ids[order]
+            v = vals.view(-1, self.num_col)[order]
+            uids, inv = torch.unique_consecutive(sids, return_inverse=True)
+            agg = torch.zeros(uids.numel(), self.num_col,
+                              dtype=v.dtype, device=v.device)
+            agg.index_add_(0, inv, v)
+            local_ids, vals = uids, agg.view(-1)
+            assume_unique = True
         if self.updater_type == "adagrad":
             from ..updaters import AddOption as _AO
             opt = option or _AO()

@@ -152,6 +152,16 @@ class Zoo:
         else:
             self.rank, self.size = 0, 1
 
+        lvl = str(get_flag("log_level")).lower()
+        from .log import LogLevel
+        if lvl in ("debug", "info", "error", "fatal"):
+            log.reset_log_level(getattr(LogLevel, lvl.upper()))
+        omp = int(get_flag("omp_threads"))
+        if omp > 0:
+            # reference updater.cpp:18-19 OpenMP thread count -> the CPU
+            # fallback's intra-op pool; 0 (default) keeps torch's choice
+            torch.set_num_threads(omp)
+
         self.started = True
         log.debug(f"Zoo started: rank {self.rank}/{self.size} "
                   f"backend={self.backend} device={self.device}")

@@ -5,7 +5,7 @@ from multiverso_amd.configure import (define_flag, get_flag, parse_cmd_flags,
 def test_defaults():
     assert get_flag("updater_type") == "default"
     assert get_flag("sync") is False
-    assert get_flag("omp_threads") == 4
+    assert get_flag("omp_threads") == 0  # 0 = torch default thread pool
 
 
 def test_parse_cmd_flags_consumes_known():

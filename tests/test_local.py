@@ -225,3 +225,15 @@ def test_updater_default_options(env):
         t.add(torch.ones(4))
         got = t.get()
         assert torch.isfinite(got).all(), u
+
+
+def test_deterministic_keyed_scatter(env):
+    mv.set_flag("deterministic", True)
+    try:
+        t = mv.MatrixTable(10, 2)
+        t.add_rows([3, 3, 5], torch.tensor([[1.0, 2.0], [3.0, 4.0],
+                                            [5.0, 6.0]]))
+        assert torch.equal(t.get_rows([3, 5]),
+                           torch.tensor([[4.0, 6.0], [5.0, 6.0]]))
+    finally:
+        mv.set_flag("deterministic", False)
