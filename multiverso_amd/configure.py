@@ -90,7 +90,7 @@ def parse_cmd_flags(argv: List[str]) -> List[str]:
 def _define_core_flags() -> None:
     # Core runtime flags; parity list from SURVEY.md §5.6 / reference flags.
     define_flag("ps_role", "default", "node role: default(worker+server)|worker|server|none")
-    define_flag("ma", False, "model-average mode: skip PS, use aggregate() allreduce")
+    define_flag("ma", False, "model-average mode (reference zoo.cpp:24): aggregate() is always available here; the flag is accepted so reference launch lines work unchanged")
     define_flag("sync", False, "BSP synchronous server (vector-clocked)")
     define_flag("backup_worker_ratio", 0.0, "vestigial in reference; kept for parity")
     define_flag("updater_type", "default", "default|sgd|momentum|adagrad")

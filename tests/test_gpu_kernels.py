@@ -345,3 +345,26 @@ def test_get_noncontiguous_buffer_fallback():
     torch.cuda.synchronize()
     assert torch.allclose(out2, -2 * delta)
     mv.shutdown()
+
+
+def test_deterministic_keyed_scatter_gpu():
+    """deterministic=1: duplicate rows pre-aggregate (no atomics) and the
+    result is bitwise stable across repeats."""
+    import multiverso_amd as mv
+    mv.init()
+    mv.set_flag("deterministic", True)
+    try:
+        rows = torch.randint(0, 100, (5000,), dtype=torch.int64).cuda()
+        vals = torch.randn(5000, 8, device="cuda:0")
+        results = []
+        for _ in range(2):
+            t = mv.MatrixTable(100, 8)
+            t.add_rows(rows, vals)
+            results.append(t.get())
+        torch.cuda.synchronize()
+        assert torch.equal(results[0], results[1])
+        ref = torch.zeros(100, 8).index_add_(0, rows.cpu(), vals.cpu())
+        assert torch.allclose(results[0].cpu(), ref, rtol=1e-5, atol=1e-5)
+    finally:
+        mv.set_flag("deterministic", False)
+        mv.shutdown()
