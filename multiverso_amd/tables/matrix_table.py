@@ -174,11 +174,15 @@ class MatrixTable(Table):
             order = torch.argsort(local_ids, stable=True)
             sids = local_ids[order]
             v = vals.view(-1, self.num_col)[order]
-            uids, inv = torch.unique_consecutive(sids, return_inverse=True)
-            agg = torch.zeros(uids.numel(), self.num_col,
-                              dtype=v.dtype, device=v.device)
-            agg.index_add_(0, inv, v)
-            local_ids, vals = uids, agg.view(-1)
+            uids, counts = torch.unique_consecutive(sids,
+                                                    return_counts=True)
+            # fp64 cumsum-diff segmented sum: fixed order, no atomics
+            cs = torch.zeros(v.size(0) + 1, self.num_col,
+                             dtype=torch.float64, device=v.device)
+            torch.cumsum(v.to(torch.float64), 0, out=cs[1:])
+            ends = counts.cumsum(0)
+            agg = (cs[ends] - cs[ends - counts]).to(v.dtype)
+            local_ids, vals = uids, agg.reshape(-1)
             assume_unique = True
         if self.updater_type == "adagrad":
             from ..updaters import AddOption as _AO
