@@ -26,3 +26,48 @@ def test_shim_surface_and_roundtrip():
     m.add(torch.ones(4, 3))
     assert torch.equal(torch.as_tensor(m.get()), torch.ones(4, 3))
     mv.shutdown()
+
+
+def _c_api_symbols():
+    import re
+    hdr = open(os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "multiverso_amd", "capi",
+        "c_api.h")).read()
+    return set(re.findall(r"\b(MV_\w+)\s*\(", hdr))
+
+
+def test_lua_binding_symbols_match_c_api():
+    """The Lua FFI cdefs must only name symbols the C API exports."""
+    import re
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    lua = open(os.path.join(root, "binding", "lua", "multiverso.lua")).read()
+    used = set(re.findall(r"\b(MV_\w+)\s*\(", lua))
+    exported = _c_api_symbols()
+    assert used, "no MV_ symbols found in the Lua binding"
+    assert used <= exported, used - exported
+
+
+def test_csharp_binding_symbols_match_c_api():
+    """Every DllImport EntryPoint must exist in c_api.h."""
+    import re
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cs = open(os.path.join(root, "binding", "csharp",
+                           "Multiverso.cs")).read()
+    used = set(re.findall(r'EntryPoint = "(MV_\w+)"', cs))
+    exported = _c_api_symbols()
+    assert used, "no EntryPoints found in the C# binding"
+    assert used <= exported, used - exported
+
+
+def test_c_api_symbol_set_matches_reference():
+    """Exact reference symbol surface (c_api.h:16-54)."""
+    expected = {
+        "MV_Init", "MV_ShutDown", "MV_Barrier", "MV_NumWorkers",
+        "MV_WorkerId", "MV_ServerId",
+        "MV_NewArrayTable", "MV_GetArrayTable", "MV_AddArrayTable",
+        "MV_AddAsyncArrayTable",
+        "MV_NewMatrixTable", "MV_GetMatrixTableAll", "MV_AddMatrixTableAll",
+        "MV_AddAsyncMatrixTableAll", "MV_GetMatrixTableByRows",
+        "MV_AddMatrixTableByRows", "MV_AddAsyncMatrixTableByRows",
+    }
+    assert _c_api_symbols() == expected
