@@ -10,8 +10,10 @@ python __graft_entry__.py build
 python -c 'from multiverso_amd import capi; capi.build(verbose=True)'
 
 echo "== CPU suite (logic + single-process runtime + world_size 2/4 gloo \
-+ binding + CLI end-to-end) =="
-python -m pytest tests -q -m "not gpu"
++ binding + CLI end-to-end; GPU masked so device selection and \
+multi-process semantics match the CPU tier) =="
+HIP_VISIBLE_DEVICES="" CUDA_VISIBLE_DEVICES="" \
+    python -m pytest tests -q -m "not gpu"
 
 if python -c 'import torch,sys; sys.exit(0 if torch.cuda.is_available() else 1)'; then
   echo "== GPU suite (kernel numerics vs fp32 torch references, table ops, \
