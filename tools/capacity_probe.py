@@ -1,11 +1,11 @@
-"""HBM-capacity / 64-bit-indexing probe: a 100 GB MatrixTable shard on
-one MI355X (25e9 fp32 elements — beyond int32 indexing), whole-table Add
+"""HBM-capacity / 64-bit-indexing probe: a 133 GiB MatrixTable shard on
+one MI355X (35.8e9 fp32 elements — beyond int32 indexing), whole-table Add
 through the fused SGD kernel, then row-keyed Get/Add at the far end of
 the table. Validates the "server shards resident in 288 GB HBM3E, no
 host staging" design point at a size the reference
 could only reach with 24+ machines.
 
-Run: python tools/capacity_probe.py  (needs a GPU with >= 210 GB free)
+Run: python tools/capacity_probe.py  (needs a GPU with >= 270 GB free)
 """
 
 import os
@@ -21,7 +21,7 @@ def main():
     assert torch.cuda.is_available(), "capacity probe needs the GPU"
     import multiverso_amd as mv
     mv.init()
-    rows, cols = 200_000_000, 128      # 25.6e9 elements, 102.4 GB fp32
+    rows, cols = 280_000_000, 128      # 35.8e9 elements, 143 GB fp32
     t = mv.MatrixTable(rows, cols, updater_type="sgd")
     n = rows * cols
     print(f"table: {rows}x{cols} = {n/1e9:.1f}G elements "
@@ -30,7 +30,8 @@ def main():
     delta = torch.ones(rows, cols, device="cuda:0")
     torch.cuda.synchronize()
     t0 = time.perf_counter()
-    t.add(delta).wait()
+    t.add(delta)
+    t.flush()      # materialize the (deferred) add: timed region is real
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     print(f"whole-table Add (sgd): {dt*1e3:.1f} ms = "
