@@ -125,6 +125,11 @@ class MatrixTable(Table):
 
     def add(self, delta: torch.Tensor, option: Optional[AddOption] = None,
             async_op: bool = False) -> Handle:
+        """Whole-table Add. Single-rank GPU adds DEFER until the next
+        table op (fusing with an immediately following Get); every public
+        read applies the pending add first, and in-place mutation of
+        ``delta`` before then is a loud error. Use ``flush()`` to force
+        materialization (e.g. before timing the update itself)."""
         CHECK(delta.numel() == self.num_row * self.num_col,
               "Add delta size mismatch")
         delta = delta.to(self.device, self.dtype).contiguous().view(-1)
