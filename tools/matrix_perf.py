@@ -23,6 +23,9 @@ def main():
     p.add_argument("--rows", type=int, default=1_000_000)
     p.add_argument("--cols", type=int, default=50)
     p.add_argument("--iters", type=int, default=3)
+    p.add_argument("--sparse", action="store_true",
+                   help="TestSparsePerf equivalent: SparseMatrixTable "
+                        "stale-filtered get_into instead of dense get")
     args = p.parse_args()
 
     import multiverso_amd as mv
@@ -31,7 +34,11 @@ def main():
     if device.type != "cuda":
         args.rows = min(args.rows, 50_000)
 
-    t = mv.MatrixTable(args.rows, args.cols, updater_type="default")
+    if args.sparse:
+        t = mv.SparseMatrixTable(args.rows, args.cols,
+                                 updater_type="default")
+    else:
+        t = mv.MatrixTable(args.rows, args.cols, updater_type="default")
     out = torch.empty(args.rows, args.cols, device=device)
 
     def timed(fn):
@@ -51,14 +58,31 @@ def main():
         k = args.rows * pct // 100
         ids = rows_all[torch.randperm(args.rows, device=device)[:k]]
         vals = torch.rand(k, args.cols, device=device)
-        get_before = timed(lambda: t.get(out=out))
-        add_ms = timed(lambda: t.add_rows(ids, vals))
-        get_after = timed(lambda: t.get(out=out))
-        results.append((pct, add_ms, get_before, get_after))
-        if mv.rank() == 0:
-            print(f"add {pct:3d}% rows ({k}): add {add_ms:8.2f} ms | "
-                  f"get-all before {get_before:8.2f} ms after "
-                  f"{get_after:8.2f} ms", flush=True)
+        if args.sparse:
+            got = [0]
+
+            def sget():
+                got[0] = t.get_into(out)
+
+            get_before = timed(sget)
+            n_before = got[0]
+            add_ms = timed(lambda: t.add_rows(ids, vals))
+            get_after = timed(sget)
+            results.append((pct, add_ms, get_before, get_after))
+            if mv.rank() == 0:
+                print(f"add {pct:3d}% rows ({k}): add {add_ms:8.2f} ms | "
+                      f"stale-get before {get_before:8.2f} ms "
+                      f"({n_before} rows) after {get_after:8.2f} ms "
+                      f"({got[0]} rows)", flush=True)
+        else:
+            get_before = timed(lambda: t.get(out=out))
+            add_ms = timed(lambda: t.add_rows(ids, vals))
+            get_after = timed(lambda: t.get(out=out))
+            results.append((pct, add_ms, get_before, get_after))
+            if mv.rank() == 0:
+                print(f"add {pct:3d}% rows ({k}): add {add_ms:8.2f} ms | "
+                      f"get-all before {get_before:8.2f} ms after "
+                      f"{get_after:8.2f} ms", flush=True)
 
     if mv.rank() == 0:
         print(mv.Dashboard.display())
