@@ -219,3 +219,19 @@ def _w2v_uneven_blocks(rank, world):
 def test_w2v_uneven_blocks_dist():
     from conftest import run_dist
     run_dist(_w2v_uneven_blocks, 2)
+
+
+def test_huffman_exact_reference_layout():
+    """Exact point/code arrays for a hand-traced run of the reference's
+    two-queue construction (huffman_encoder.cpp:87-188) on counts
+    [4,2,1,1]: code = root->leaf branch labels, point = inner-node path
+    (root = V-2 first, leaf's parent last)."""
+    enc = HuffmanEncoder()
+    enc.build_from_term_frequency([4, 2, 1, 1])
+    expect = [([2], [1]),
+              ([2, 1], [0, 1]),
+              ([2, 1, 0], [0, 0, 1]),
+              ([2, 1, 0], [0, 0, 0])]
+    for i, (point, code) in enumerate(expect):
+        assert enc.labels[i].point == point, (i, enc.labels[i].point)
+        assert enc.labels[i].code == code, (i, enc.labels[i].code)
