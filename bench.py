@@ -38,6 +38,8 @@ def parse_args():
     p.add_argument("--dim", type=int, default=200)
     p.add_argument("--block-words", dest="block_words", type=int,
                    default=500_000)
+    p.add_argument("--use-adagrad", dest="use_adagrad", action="store_true",
+                   help="wordembedding: per-element AdaGrad mode")
     return p.parse_args()
 
 

@@ -30,6 +30,8 @@ def run_bench(args):
         vocab, block_words = min(vocab, 20_000), min(block_words, 20_000)
 
     opt = WordEmbeddingOption(embedding_size=dim, window=5, negative_num=5,
+                              use_adagrad=getattr(args, "use_adagrad",
+                                                  False),
                               total_words=block_words * args.steps * n,
                               seed=17)
     counts = zipf_counts(vocab, opt.total_words)
@@ -83,7 +85,8 @@ def run_bench(args):
             "dtype": "fp32",
             "data": "synthetic",
             "config": {
-                "model": f"word2vec skip-gram dim={dim} vocab={vocab} neg=5",
+                "model": f"word2vec skip-gram dim={dim} vocab={vocab} neg=5"
+                         + (" adagrad" if getattr(args, 'use_adagrad', False) else ''),
                 "global_batch": n * block_words,
                 "seq_len": None,
                 "parallelism": f"ps-sharded dp{n} (row all-to-all over xGMI)",
