@@ -15,6 +15,9 @@ Capability parity with the reference updater family (selected by the
            intended accumulate semantics. In the collective data plane the
            reduce-scattered delta is the sum over workers, so there is one
            accumulator per shard rather than one per (worker, shard).
+- dcasgd / dcasgda: delay-compensated ASGD (selected at updater.cpp:51-54
+           from a submodule absent in the snapshot; math from Zheng et
+           al., ICML 2017) with per-worker backup rows.
 
 ``AddOption`` keeps the reference's 20-byte wire envelope
 (include/multiverso/updater/updater.h:10-70) for C-API parity.
