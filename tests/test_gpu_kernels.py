@@ -368,3 +368,25 @@ def test_deterministic_keyed_scatter_gpu():
     finally:
         mv.set_flag("deterministic", False)
         mv.shutdown()
+
+
+def test_logreg_local_mode_gpu():
+    """use_ps=false LocalModel on the GPU (K13-K15 via scatter kernels)."""
+    import multiverso_amd as mv
+    from multiverso_amd.apps.logreg.config import LogRegConfig
+    from multiverso_amd.apps.logreg.model import LocalModel
+    from multiverso_amd.apps.logreg.reader import synthetic_batches
+    mv.init()
+    cfg = LogRegConfig(input_size=100_000, output_size=1,
+                       objective_type="sigmoid", updater_type="sgd",
+                       learning_rate=0.1, minibatch_size=512,
+                       sparse=True)
+    m = LocalModel(cfg, device=torch.device("cuda:0"))
+    batches, w_true = synthetic_batches(cfg.input_size, 30, 512, nnz=16,
+                                        seed=3, device="cuda:0")
+    losses = [m.update(b) for b in batches]
+    torch.cuda.synchronize()
+    assert losses[-1] < losses[0], (losses[0], losses[-1])
+    p = m.predict(batches[0])
+    assert p.shape == (512, 1) and torch.isfinite(p).all()
+    mv.shutdown()
