@@ -390,3 +390,29 @@ def test_logreg_local_mode_gpu():
     p = m.predict(batches[0])
     assert p.shape == (512, 1) and torch.isfinite(p).all()
     mv.shutdown()
+
+
+def test_torch_ext_gpu():
+    """Param-manager and shared-tensor ASGD sync on GPU (exercises the
+    fused Add+Get through the binding protocols)."""
+    import multiverso_amd as mv
+    from multiverso_amd.torch_ext import MVTorchParamManager, MVSharedTensor
+    mv.init()
+    lin = torch.nn.Linear(32, 4).cuda()
+    mgr = MVTorchParamManager(lin)
+    before = [p.detach().clone() for p in lin.parameters()]
+    with torch.no_grad():
+        for p in lin.parameters():
+            p.add_(0.5)
+    mgr.sync_all_param()
+    torch.cuda.synchronize()
+    for p, b in zip(lin.parameters(), before):
+        assert torch.allclose(p.detach(), b + 0.5, atol=1e-6)
+
+    t = torch.zeros(64, device="cuda:0")
+    sv = MVSharedTensor(t)
+    t.add_(2.0)
+    sv.mv_sync()
+    torch.cuda.synchronize()
+    assert torch.allclose(t, torch.full((64,), 2.0, device="cuda:0"))
+    mv.shutdown()
