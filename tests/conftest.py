@@ -31,6 +31,11 @@ def free_port() -> int:
 
 
 def _dist_entry(rank: int, fn, world_size: int, port: int, backend: str):
+    if backend == "gloo":
+        # these are CPU-tier tests: mask any GPU so world_size ranks on a
+        # 1-GPU host don't map LOCAL_RANK -> missing cuda devices
+        os.environ["HIP_VISIBLE_DEVICES"] = ""
+        os.environ["CUDA_VISIBLE_DEVICES"] = ""
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     os.environ["RANK"] = str(rank)
