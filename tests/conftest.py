@@ -10,6 +10,11 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: test requires an MI355X GPU (run with -m gpu)")
+    # The CPU tier must behave identically on GPU-equipped hosts:
+    # -m "not gpu" masks the devices before torch can initialize them.
+    if "not gpu" in (config.getoption("-m") or ""):
+        os.environ["HIP_VISIBLE_DEVICES"] = ""
+        os.environ["CUDA_VISIBLE_DEVICES"] = ""
 
 
 def pytest_collection_modifyitems(config, items):
