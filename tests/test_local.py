@@ -237,3 +237,19 @@ def test_deterministic_keyed_scatter(env):
                            torch.tensor([[4.0, 6.0], [5.0, 6.0]]))
     finally:
         mv.set_flag("deterministic", False)
+
+
+def test_lifecycle_robustness():
+    # double init is idempotent; shutdown twice is safe; re-init gives a
+    # fresh table registry (reference MV_ShutDown + re-Init semantics)
+    mv.init()
+    mv.init()
+    t = mv.ArrayTable(4)
+    t.add(torch.ones(4))
+    mv.shutdown()
+    mv.shutdown()
+    mv.init()
+    t2 = mv.ArrayTable(4)
+    assert torch.equal(t2.get(), torch.zeros(4))
+    mv.shutdown()
+    mv.barrier()   # no-op outside a session
