@@ -76,3 +76,30 @@ def test_logreg_cli(tmp_path):
     import re
     m = re.search(r"\((0\.\d+)\)", r.stdout + r.stderr)
     assert m and float(m.group(1)) > 0.55, (r.stdout + r.stderr)[-400:]
+
+
+def test_logreg_cli_softmax(tmp_path):
+    """Multiclass softmax through the CLI (separable 5-class synthetic)."""
+    import re
+    train = tmp_path / "sm.txt"
+    rng = random.Random(9)
+    with open(train, "w") as f:
+        for i in range(400):
+            label = i % 5
+            keys = sorted(rng.sample(range(40 * label, 40 * label + 40), 8))
+            f.write(f"{label} " + " ".join(f"{k}:1" for k in keys) + "\n")
+    cfg = tmp_path / "sm.config"
+    model = tmp_path / "sm.bin"
+    cfg.write_text(
+        "input_size=200\noutput_size=5\nobjective_type=softmax\n"
+        "updater_type=sgd\nlearning_rate=0.1\nminibatch_size=32\n"
+        "train_epoch=3\nreader_type=default\nuse_ps=true\n"
+        f"sync_frequency=2\ntrain_file={train}\ntest_file={train}\n"
+        f"output_model_file={model}\n")
+    r = subprocess.run(
+        [sys.executable, "-m", "multiverso_amd.apps.logreg.main", str(cfg)],
+        cwd=ROOT, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    assert model.stat().st_size == 200 * 5 * 4
+    m = re.search(r"\((0\.\d+|1\.0000)\)", r.stdout + r.stderr)
+    assert m and float(m.group(1)) > 0.9, (r.stdout + r.stderr)[-300:]
