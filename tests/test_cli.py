@@ -103,3 +103,31 @@ def test_logreg_cli_softmax(tmp_path):
     assert model.stat().st_size == 200 * 5 * 4
     m = re.search(r"\((0\.\d+|1\.0000)\)", r.stdout + r.stderr)
     assert m and float(m.group(1)) > 0.9, (r.stdout + r.stderr)[-300:]
+
+
+def test_logreg_cli_ftrl(tmp_path):
+    """FTRL objective+updater through the CLI on a separable problem."""
+    import re
+    train = tmp_path / "ftrl.txt"
+    rng = random.Random(4)
+    with open(train, "w") as f:
+        for i in range(400):
+            pos = i % 2
+            lo = 0 if pos else 100
+            keys = sorted(rng.sample(range(lo, lo + 100), 8))
+            f.write(f"{pos} " + " ".join(f"{k}:1" for k in keys) + "\n")
+    cfg = tmp_path / "f.config"
+    model = tmp_path / "f.bin"
+    cfg.write_text(
+        "input_size=200\noutput_size=1\nobjective_type=ftrl\n"
+        "updater_type=ftrl\nalpha=0.1\nbeta=1\nlambda1=0.01\nlambda2=0.01\n"
+        "minibatch_size=32\ntrain_epoch=3\nreader_type=default\nuse_ps=true\n"
+        f"sync_frequency=2\ntrain_file={train}\ntest_file={train}\n"
+        f"output_model_file={model}\n")
+    r = subprocess.run(
+        [sys.executable, "-m", "multiverso_amd.apps.logreg.main", str(cfg)],
+        cwd=ROOT, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    import re as _re
+    m = _re.search(r"\((0\.\d+|1\.0000)\)", r.stdout + r.stderr)
+    assert m and float(m.group(1)) > 0.9, (r.stdout + r.stderr)[-300:]
