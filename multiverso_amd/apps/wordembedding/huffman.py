@@ -70,3 +70,32 @@ class HuffmanEncoder:
 
     def get_label_info(self, word_id: int) -> HuffLabelInfo:
         return self.labels[word_id]
+
+
+    # ---- file io (huffman_encoder.cpp:9-85 format) ----
+    def save_to_file(self, path: str, words: List[str]) -> None:
+        """Save2File: header = vocab size; per line
+        "word codelen code... point..."."""
+        with open(path, "w") as f:
+            f.write(f"{len(self.labels)}\n")
+            for i, info in enumerate(self.labels):
+                parts = [words[i], str(len(info.code))]
+                parts += [str(c) for c in info.code]
+                parts += [str(pt) for pt in info.point]
+                f.write(" ".join(parts) + "\n")
+
+    def load_from_file(self, path: str) -> List[str]:
+        """RecoverFromFile: returns the word list; labels loaded in
+        file order."""
+        with open(path) as f:
+            n = int(f.readline().split()[0])
+            words: List[str] = []
+            self.labels = []
+            for _ in range(n):
+                parts = f.readline().split()
+                w, codelen = parts[0], int(parts[1])
+                code = [int(x) for x in parts[2:2 + codelen]]
+                point = [int(x) for x in parts[2 + codelen:2 + 2 * codelen]]
+                words.append(w)
+                self.labels.append(HuffLabelInfo(point, code))
+        return words

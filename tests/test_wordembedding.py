@@ -235,3 +235,18 @@ def test_huffman_exact_reference_layout():
     for i, (point, code) in enumerate(expect):
         assert enc.labels[i].point == point, (i, enc.labels[i].point)
         assert enc.labels[i].code == code, (i, enc.labels[i].code)
+
+
+def test_huffman_file_roundtrip(tmp_path):
+    """Save2File/RecoverFromFile format (huffman_encoder.cpp:9-85)."""
+    enc = HuffmanEncoder()
+    enc.build_from_term_frequency([9, 5, 3, 2, 1])
+    p = str(tmp_path / "huff.txt")
+    enc.save_to_file(p, ["a", "b", "c", "d", "e"])
+    enc2 = HuffmanEncoder()
+    words = enc2.load_from_file(p)
+    assert words == ["a", "b", "c", "d", "e"]
+    for x, y in zip(enc.labels, enc2.labels):
+        assert x.code == y.code and x.point == y.point
+    first = open(p).read().splitlines()
+    assert first[0] == "5" and first[1].startswith("a ")
