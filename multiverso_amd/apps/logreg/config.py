@@ -58,7 +58,8 @@ class LogRegConfig:
                 key, _, val = line.partition("=")
                 key, val = key.strip(), val.strip()
                 if not hasattr(cfg, key):
-                    print(f"[logreg] unknown config key '{key}' ignored")
+                    from multiverso_amd.log import log
+                    log.error(f"[logreg] unknown config key '{key}' ignored")
                     continue
                 cur = getattr(cfg, key)
                 if isinstance(cur, bool):
