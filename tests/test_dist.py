@@ -186,5 +186,10 @@ def test_array_uneven_shards_ws4():
     run_dist(_array_uneven, 4)
 
 
+def test_array_uneven_shards_ws3():
+    # odd world: 11 elements shard as 3+3+5 (remainder on the last server)
+    run_dist(_array_uneven, 3)
+
+
 def test_matrix_whole_and_rows_ws4():
     run_dist(_matrix_whole_and_rows, 4)
