@@ -46,3 +46,28 @@ def test_sparse_filter_roundtrip_any_density(n, density, seed):
     assert torch.equal(out, v)
     if comp:  # compression only when it actually shrinks the payload
         assert payload.numel() < n
+
+
+@given(counts=st.lists(st.integers(min_value=1, max_value=10_000),
+                       min_size=2, max_size=120),
+       )
+@settings(max_examples=50, deadline=None)
+def test_huffman_invariants(counts):
+    from multiverso_amd.apps.wordembedding.huffman import HuffmanEncoder
+    enc = HuffmanEncoder()
+    enc.build_from_term_frequency(counts)
+    v = len(counts)
+    codes = ["".join(map(str, l.code)) for l in enc.labels]
+    # prefix-free
+    sc = sorted(codes)
+    for a, b in zip(sc, sc[1:]):
+        assert not b.startswith(a)
+    # weighted path length is optimal-ish: shorter codes for higher counts
+    # (weak: the max-count word's code is no longer than the min-count's)
+    import numpy as np
+    mx, mn = int(np.argmax(counts)), int(np.argmin(counts))
+    assert len(codes[mx]) <= len(codes[mn])
+    for l in enc.labels:
+        assert len(l.point) == len(l.code) >= 1
+        assert all(0 <= p < v - 1 for p in l.point)
+        assert l.point[0] == v - 2  # root first (reference layout)
