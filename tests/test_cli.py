@@ -131,3 +131,43 @@ def test_logreg_cli_ftrl(tmp_path):
     import re as _re
     m = _re.search(r"\((0\.\d+|1\.0000)\)", r.stdout + r.stderr)
     assert m and float(m.group(1)) > 0.9, (r.stdout + r.stderr)[-300:]
+
+
+def test_embedding_server(tmp_path):
+    """Serving example: word2vec file -> PS MatrixTable -> HTTP lookups
+    (fastapi TestClient, no network)."""
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        "serve_embeddings", os.path.join(ROOT, "examples",
+                                         "serve_embeddings.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+
+    # tiny text-format embedding file
+    p = tmp_path / "emb.txt"
+    with open(p, "w") as f:
+        f.write("4 3\n")
+        f.write("king 1.0 0.0 0.0\n")
+        f.write("queen 0.9 0.1 0.0\n")
+        f.write("apple 0.0 1.0 0.0\n")
+        f.write("pear 0.0 0.9 0.1\n")
+    words, vecs = mod.load_word2vec(str(p))
+    assert words == ["king", "queen", "apple", "pear"]
+    assert vecs.shape == (4, 3)
+
+    import multiverso_amd as mv
+    mv.init()
+    table = mv.MatrixTable(4, 3)
+    table.add(vecs)
+    table.flush()
+    from fastapi.testclient import TestClient
+    client = TestClient(mod.build_app(table, words, 3))
+    assert client.get("/healthz").json()["vocab"] == 4
+    v = client.get("/vec/king").json()
+    assert v["vector"] == [1.0, 0.0, 0.0]
+    nn = client.get("/nn/king?k=1").json()
+    assert nn["neighbors"][0]["word"] == "queen"
+    nn2 = client.get("/nn/apple?k=1").json()
+    assert nn2["neighbors"][0]["word"] == "pear"
+    assert client.get("/vec/zzz").status_code == 404
+    mv.shutdown()
