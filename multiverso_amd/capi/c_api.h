@@ -15,23 +15,42 @@
 extern "C" {
 #endif
 
+/* Opaque table reference returned by the MV_New* constructors. The
+ * handle owns a Python-side table object inside the embedded runtime;
+ * it stays valid until MV_ShutDown. */
 typedef void* TableHandler;
 
+/* Runtime lifecycle. MV_Init boots the embedded MI355X runtime in this
+ * process (rendezvous from the launcher environment, or from an argv
+ * of "-key=value" flags such as -sync=true); MV_ShutDown drains pending
+ * table work and tears the process group down; MV_Barrier blocks until
+ * every rank arrives. */
 DllExport void MV_Init(int* argc, char* argv[]);
 DllExport void MV_ShutDown();
 DllExport void MV_Barrier();
+
+/* Topology queries: every rank is worker AND server in this design, so
+ * NumWorkers == world size and WorkerId == ServerId == rank. */
 DllExport int MV_NumWorkers();
 DllExport int MV_WorkerId();
 DllExport int MV_ServerId();
 
-/* Array Table */
+/* 1-D float table, contiguous-sharded across ranks. Get fills `data`
+ * (length `size` floats) with the whole table via an all-gather of the
+ * HBM-resident shards; Add applies `data` as a whole-table delta via a
+ * reduce-scatter plus the server-side updater kernel. The Async add
+ * returns once the payload is snapshotted; ordering against later Gets
+ * is preserved. */
 DllExport void MV_NewArrayTable(int size, TableHandler* out);
 DllExport void MV_GetArrayTable(TableHandler handler, float* data, int size);
 DllExport void MV_AddArrayTable(TableHandler handler, float* data, int size);
 DllExport void MV_AddAsyncArrayTable(TableHandler handler, float* data,
                                      int size);
 
-/* Matrix Table */
+/* 2-D float table, row-sharded. The *All calls move the whole matrix
+ * (size == num_row * num_col floats, row-major); the *ByRows calls move
+ * the `row_ids_n` rows listed in `row_ids` (size == row_ids_n * num_col
+ * floats), exchanged with their owning ranks as a batched all-to-all. */
 DllExport void MV_NewMatrixTable(int num_row, int num_col, TableHandler* out);
 DllExport void MV_GetMatrixTableAll(TableHandler handler, float* data,
                                     int size);
