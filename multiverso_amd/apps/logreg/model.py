@@ -121,9 +121,16 @@ class PSModel:
         the pulled snapshot (within-chunk updates visible locally), push
         the summed delta (ps_model.cpp:172-203)."""
         batches = [b.to(self.device) for b in batches]
-        union = (torch.unique(torch.cat([b.keys for b in batches]))
-                 if batches else
-                 torch.empty(0, dtype=torch.int64, device=self.device))
+        if batches:
+            cat_keys = torch.cat([b.keys for b in batches])
+            if self.cfg.input_size <= (1 << 31) - 1:
+                # keys fit int32: the dedup sort runs at half the bytes
+                union = torch.unique(cat_keys.to(torch.int32)).to(
+                    torch.int64)
+            else:
+                union = torch.unique(cat_keys)
+        else:
+            union = torch.empty(0, dtype=torch.int64, device=self.device)
         pulled = self.table.get_rows(union)
         local = pulled.clone()
         total_loss = 0.0
