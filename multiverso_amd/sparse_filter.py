@@ -24,31 +24,39 @@ from .dashboard import Dashboard
 
 def try_compress(values: torch.Tensor) -> Tuple[bool, torch.Tensor]:
     """If >50% of ``values`` are zero, return (True, packed) where packed =
-    [count | idx...| val...] in one flat float32 tensor (indices stored as
-    float bit-pattern-free ints — exact for payloads < 2^24 elements, which
-    row exchanges are). Else (False, values)."""
+    [count | idx...| val...] in one flat float32 tensor. The count and the
+    indices are int32 values REINTERPRETED as float32 bits (torch .view),
+    not converted — exact for any payload up to 2^31 elements; larger
+    payloads pass through dense (they would overflow int32 indexing)."""
     flat = values.reshape(-1)
-    nz = torch.nonzero(flat, as_tuple=False).reshape(-1)
     m = Dashboard.get("sparse_filter.bytes_in")
     m.count += 1
     m.elapsed_ms += flat.numel() * 4
+    if flat.numel() >= (1 << 31):
+        Dashboard.get("sparse_filter.bytes_out").elapsed_ms += flat.numel() * 4
+        return False, flat
+    nz = torch.nonzero(flat, as_tuple=False).reshape(-1)
     if nz.numel() * 2 + 1 >= flat.numel():
         Dashboard.get("sparse_filter.bytes_out").elapsed_ms += flat.numel() * 4
         return False, flat
     packed = torch.empty(1 + 2 * nz.numel(), dtype=torch.float32,
                          device=flat.device)
-    packed[0] = float(nz.numel())
-    packed[1:1 + nz.numel()] = nz.float()
+    packed[0] = torch.tensor([nz.numel()],
+                             dtype=torch.int32).view(torch.float32)[0]
+    packed[1:1 + nz.numel()] = nz.to(torch.int32).view(torch.float32)
     packed[1 + nz.numel():] = flat[nz]
     Dashboard.get("sparse_filter.bytes_out").elapsed_ms += packed.numel() * 4
     return True, packed
 
 
 def decompress(packed: torch.Tensor, out_numel: int) -> torch.Tensor:
-    """Inverse of try_compress for a compressed payload."""
-    k = int(packed[0].item())
+    """Inverse of try_compress for a compressed payload. The nonzero
+    count is derived from the payload size (numel = 1 + 2k), so no
+    device read of the count header is needed — the header stays on the
+    wire for format parity and as a consistency check on CPU."""
+    k = (packed.numel() - 1) // 2
     out = torch.zeros(out_numel, dtype=torch.float32, device=packed.device)
-    idx = packed[1:1 + k].long()
+    idx = packed[1:1 + k].view(torch.int32).long()
     out[idx] = packed[1 + k:1 + 2 * k]
     return out
 

@@ -69,7 +69,10 @@ class ArrayTable(Table):
                 out = torch.empty(self.size, dtype=self.dtype,
                                   device=self.device)
             CHECK(out.numel() == self.size, "Get buffer size mismatch")
-            if out.is_contiguous():
+            # fused update+copy kernel needs a GPU fp32 contiguous target
+            # (see matrix_table.get); anything else materializes the Add
+            if (out.is_contiguous() and out.is_cuda
+                    and out.dtype == self.dtype):
                 with monitor("server.update"):
                     self.updater.update_and_copy(d[0], d[1], out.view(-1))
                 return out

@@ -91,11 +91,14 @@ class MatrixTable(Table):
                                   dtype=self.dtype, device=self.device)
             CHECK(out.numel() == self.num_row * self.num_col,
                   "Get buffer size mismatch")
-            if out.is_contiguous():
+            # fused update+copy kernel needs a GPU fp32 contiguous target
+            # (the deferral itself only happens on GPU shards); any other
+            # out materializes the Add and falls through to the copy path
+            if (out.is_contiguous() and out.is_cuda
+                    and out.dtype == self.dtype):
                 with monitor("server.update"):
                     self.updater.update_and_copy(d[0], d[1], out.view(-1))
                 return out
-            # non-contiguous Get buffer: materialize, fall through
             with monitor("server.update"):
                 self.updater.update(d[0], d[1])
         self.flush()
@@ -225,26 +228,24 @@ class MatrixTable(Table):
                                   vals.view(-1, self.num_col) * sign)
 
     def get_rows(self, row_ids) -> torch.Tensor:
-        """Row-subset Get: returns [len(row_ids), num_col] in caller order."""
+        """Row-subset Get: returns [len(row_ids), num_col] in caller order.
+
+        ``row_ids`` is planned on the host (CPU ids are sync-free; device
+        ids cost one D2H) and the split sizes ride the gloo control lane —
+        no device sync between launch and the value all-to-all."""
         self.flush()
-        ids = torch.as_tensor(row_ids, dtype=torch.int64, device=self.device)
+        ids = torch.as_tensor(row_ids, dtype=torch.int64)
         with monitor("worker.get_rows"):
-            in_ids, _, recv_sizes, order = all_to_all_rows(
-                ids, None, self.spec, self.num_col)
+            in_ids, _, recv_sizes, order, send_sizes = all_to_all_rows(
+                ids, None, self.spec, self.num_col, device=self.device)
             # serve: gather requested rows from my shard
             local = self._local_rows_of(in_ids)
             served = self._gather_local(local)
-            # reply: route rows back to the requesters
+            # reply: route rows back to the requesters (what I received I
+            # serve back — sizes swap roles, nothing recomputed)
             if self.zoo.size > 1:
-                import torch.distributed as dist
-                send_back = recv_sizes
-                # sizes we originally sent per rank = how many of our ids
-                # went to each owner
-                owners = torch.div(ids, max(self.spec.total // self.spec.n, 1),
-                                   rounding_mode="floor").clamp_(max=self.spec.n - 1)
-                sent = torch.bincount(owners, minlength=self.spec.n).tolist()
-                flat = all_to_all_values(served.view(-1), send_back, sent,
-                                         self.num_col)
+                flat = all_to_all_values(served.view(-1), recv_sizes,
+                                         send_sizes, self.num_col)
                 got = flat.view(-1, self.num_col)
             else:
                 got = served
@@ -263,13 +264,14 @@ class MatrixTable(Table):
         scatter then skips atomics; with multiple ranks incoming ids can
         still collide across ranks, so atomics stay."""
         self.flush()   # a deferred whole-table Add must land first
-        ids = torch.as_tensor(row_ids, dtype=torch.int64, device=self.device)
+        ids = torch.as_tensor(row_ids, dtype=torch.int64)
         vals = values.to(self.device, self.dtype).contiguous()
         CHECK(vals.numel() == ids.numel() * self.num_col,
               "add_rows values size mismatch")
         with monitor("worker.add_rows"):
-            in_ids, in_vals, _, _ = all_to_all_rows(
-                ids, vals.view(-1), self.spec, self.num_col)
+            in_ids, in_vals, _, _, _ = all_to_all_rows(
+                ids, vals.view(-1), self.spec, self.num_col,
+                device=self.device)
             if in_ids.numel():
                 local = self._local_rows_of(in_ids)
                 with monitor("server.update_rows"):

@@ -40,7 +40,10 @@ class SparseMatrixTable(MatrixTable):
         # sparse_matrix_table.cpp:148-153): per-destination segments are
         # compressed when >50% zero; byte accounting on the Dashboard.
         from ..configure import get_flag
-        self.use_sparse_filter = bool(get_flag("sparse_filter"))
+        # the filter's wire format is float32; non-f32 tables go dense
+        # (silent precision loss otherwise — ADVICE r1)
+        self.use_sparse_filter = (bool(get_flag("sparse_filter"))
+                                  and dtype == torch.float32)
         nw = self.zoo.num_workers
         # up_to_date[w, r] == True -> worker w has the current row r
         self.up_to_date = torch.zeros(nw, self.local_rows, dtype=torch.bool,
@@ -63,12 +66,13 @@ class SparseMatrixTable(MatrixTable):
     def add_rows(self, row_ids, values, option: Optional[AddOption] = None,
                  source_worker: Optional[int] = None) -> None:
         self.flush()   # a deferred whole-table Add must land first
-        ids = torch.as_tensor(row_ids, dtype=torch.int64, device=self.device)
+        ids = torch.as_tensor(row_ids, dtype=torch.int64)
         vals = values.to(self.device, self.dtype).contiguous()
         from ..comm import all_to_all_rows
         with monitor("worker.add_rows"):
-            in_ids, in_vals, recv_sizes, _ = all_to_all_rows(
-                ids, vals.view(-1), self.spec, self.num_col)
+            in_ids, in_vals, recv_sizes, _, _ = all_to_all_rows(
+                ids, vals.view(-1), self.spec, self.num_col,
+                device=self.device)
             if in_ids.numel():
                 local = self._local_rows_of(in_ids)
                 with monitor("server.update_rows"):
@@ -84,37 +88,33 @@ class SparseMatrixTable(MatrixTable):
                                  send_rows, recv_rows) -> torch.Tensor:
         """Value half of the stale-row exchange with SparseFilter applied
         per destination segment (FilterIn on send, FilterOut on receive —
-        quantization_util.h:95-154 semantics over the collective)."""
+        quantization_util.h:95-154 semantics over the collective). The
+        per-segment payload sizes AND compressed flags travel in ONE
+        control-lane message, so the receiver never reads per-segment
+        device headers (no .item() syncs — ADVICE r1)."""
         from .. import sparse_filter as sf
+        from ..comm import exchange_size_rows
         device = self.device
-        packed_parts, packed_sizes = [], []
+        packed_parts, packed_sizes, flags = [], [], []
         off = 0
-        one = torch.ones(1, dtype=torch.float32, device=device)
-        zero = torch.zeros(1, dtype=torch.float32, device=device)
         for n in send_rows:
             seg = served[off:off + n].reshape(-1)
             off += n
             payload, compressed = sf.filter_in(seg)
-            # self-describing wire: 1-float header = compressed flag
-            payload = torch.cat([one if compressed else zero,
-                                 payload.float()])
             packed_parts.append(payload)
             packed_sizes.append(payload.numel())
+            flags.append(1 if compressed else 0)
         send_buf = (torch.cat(packed_parts) if packed_parts else
-                    torch.empty(0, device=device))
-        scnt = torch.tensor(packed_sizes, dtype=torch.int64, device=device)
-        rcnt = torch.empty_like(scnt)
-        dist.all_to_all_single(rcnt, scnt)
-        rsizes = rcnt.tolist()
+                    torch.empty(0, dtype=torch.float32, device=device))
+        rsizes, rflags = exchange_size_rows([packed_sizes, flags])
         recv_buf = torch.empty(sum(rsizes), dtype=torch.float32,
                                device=device)
         dist.all_to_all_single(recv_buf, send_buf, rsizes, packed_sizes)
         outs, off = [], 0
-        for n, sz in zip(recv_rows, rsizes):
+        for n, sz, fl in zip(recv_rows, rsizes, rflags):
             payload = recv_buf[off:off + sz]
             off += sz
-            compressed = bool(payload[0] > 0.5)
-            outs.append(sf.filter_out(payload[1:], compressed,
+            outs.append(sf.filter_out(payload, bool(fl),
                                       n * self.num_col))
         return (torch.cat(outs) if outs else
                 torch.empty(0, dtype=self.dtype, device=device))
@@ -140,12 +140,9 @@ class SparseMatrixTable(MatrixTable):
             stale_lists = [(~self.up_to_date[w]).nonzero().reshape(-1)
                            for w in range(nw)]
             send_rows = [int(s.numel()) for s in stale_lists]
-            # 2) exchange counts, then ids+values
-            cnt = torch.tensor(send_rows, dtype=torch.int64,
-                               device=self.device)
-            rcnt = torch.empty(nw, dtype=torch.int64, device=self.device)
-            dist.all_to_all_single(rcnt, cnt)
-            recv_rows = rcnt.tolist()
+            # 2) counts ride the control lane; then ids+values
+            from ..comm import exchange_sizes
+            recv_rows = exchange_sizes(send_rows)
             all_ids = torch.cat(stale_lists) + self.row_offset
             got_ids = torch.empty(sum(recv_rows), dtype=torch.int64,
                                   device=self.device)

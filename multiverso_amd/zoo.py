@@ -52,6 +52,11 @@ class Zoo:
         self._next_table_id = 0
         self._owns_pg = False
         self.backend = "none"
+        # host control lane (tiny control messages ride a
+        # CPU lane, not the xGMI data lane): a gloo subgroup that carries
+        # CPU scalars/counts even when the main backend is RCCL. None at
+        # world size 1 or when the main group is already gloo.
+        self.control_pg = None
         # explicit rendezvous (MV_NetBind/MV_NetConnect,
         # src/multiverso.cpp:58-68): rank + rank->endpoint map set before
         # init, replacing the launcher-env rendezvous the way the
@@ -149,6 +154,13 @@ class Zoo:
         if dist.is_initialized():
             self.rank = dist.get_rank()
             self.size = dist.get_world_size()
+            if self.size > 1 and self.backend == "nccl":
+                # gloo control lane alongside RCCL: CPU control scalars
+                # (uneven-stream participation flags, keyed-op split
+                # sizes) must not ride the GPU data lane — RCCL rejects
+                # CPU tensors and a count exchange on-device forces a
+                # stream sync.
+                self.control_pg = dist.new_group(backend="gloo")
         else:
             self.rank, self.size = 0, 1
 
@@ -177,6 +189,9 @@ class Zoo:
         free_tables()
         self._tables.clear()
         self._next_table_id = 0
+        if dist.is_initialized() and self.control_pg is not None:
+            dist.destroy_process_group(self.control_pg)
+        self.control_pg = None
         if self._owns_pg and dist.is_initialized() and finalize_net:
             dist.destroy_process_group()
         self.started = False
@@ -222,6 +237,12 @@ class Zoo:
         the reference's 4096-byte small-message switch
         (allreduce_engine.cpp:35), sized for 7 xGMI links instead."""
         if not dist.is_initialized():
+            return tensor
+        if (not tensor.is_cuda) and self.control_pg is not None:
+            # CPU scalar/flag aggregates under an RCCL main backend ride
+            # the gloo control lane (RCCL rejects CPU tensors)
+            dist.all_reduce(tensor, op=dist.ReduceOp.SUM,
+                            group=self.control_pg)
             return tensor
         if bucket_mb is None:
             bucket_mb = int(get_flag("bucket_mb"))

@@ -62,7 +62,8 @@ def _all_to_all_routing(rank, world):
     # rank 0 sends rows [7, 0, 4]; rank 1 sends [1, 5]
     ids = torch.tensor([7, 0, 4] if rank == 0 else [1, 5])
     vals = ids.float().repeat_interleave(2)  # unit=2
-    in_ids, in_vals, recv_sizes, order = all_to_all_rows(ids, vals, spec, 2)
+    in_ids, in_vals, recv_sizes, order, send_sizes = all_to_all_rows(
+        ids, vals, spec, 2)
     # every received id must belong to my shard, with matching values
     off, cnt = spec.range_of(rank)
     assert torch.all((in_ids >= off) & (in_ids < off + cnt)), in_ids

@@ -91,6 +91,23 @@ def test_sparse_filter_roundtrip():
     assert not comp and torch.equal(payload, dense)
 
 
+def test_sparse_filter_exact_above_2pow24():
+    """Indices are int32 bit patterns, not float values: payloads larger
+    than 2^24 elements (where float32 integers stop being exact) must
+    round-trip exactly (float-valued indices silently
+    corrupted rows past 16.7M elements)."""
+    from multiverso_amd import sparse_filter as sf
+    n = (1 << 24) + 8
+    v = torch.zeros(n)
+    hot = [0, (1 << 24) - 1, (1 << 24), (1 << 24) + 1, n - 1]
+    v[hot] = torch.tensor([1.0, 2.0, 3.0, 4.0, 5.0])
+    payload, comp = sf.filter_in(v)
+    assert comp
+    back = sf.filter_out(payload, comp, n)
+    assert torch.equal(back[hot], v[hot])
+    assert float(back.sum()) == 15.0   # nothing landed anywhere else
+
+
 # ---- sparse matrix table ----
 
 def test_sparse_matrix_local(env):
