@@ -127,19 +127,15 @@ class ArrayTable(Table):
 
     # ---- checkpoint (Store/Load, array_table.cpp:144-151) ----
     def store(self, path: str) -> None:
-        """Write whole-table raw bytes (== concatenated server shard dumps,
-        byte-identical to the reference's per-shard Store layout)."""
-        full = self.get()
-        if self.zoo.rank == 0:
-            full.cpu().numpy().tofile(path)
-        self.zoo.barrier()
+        """Write whole-table raw bytes, byte-identical to the reference's
+        concatenated per-shard Store layout. Streaming: each rank pwrites
+        its own shard at its offset — no rank-0 full-table gather."""
+        self.flush()
+        off, _ = self.spec.range_of(self.zoo.server_id)
+        self._store_shard_stream(path, self.shard, off, self.size)
 
     def load(self, path: str) -> None:
+        """Each rank reads only its own shard slice (streamed)."""
         self.flush()
-        import numpy as np
-        arr = np.fromfile(path, dtype=str(self.dtype).replace("torch.", ""))
-        CHECK(arr.size == self.size, "checkpoint size mismatch")
-        full = torch.from_numpy(arr).to(self.device)
-        off, cnt = self.spec.range_of(self.zoo.server_id)
-        self.shard.copy_(full[off:off + cnt])
-        self.zoo.barrier()
+        off, _ = self.spec.range_of(self.zoo.server_id)
+        self._load_shard_stream(path, self.shard, off, self.size)

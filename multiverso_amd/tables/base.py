@@ -17,6 +17,7 @@ observes every Add issued before it on this worker.
 
 from __future__ import annotations
 
+import os
 from typing import List, Optional
 
 import torch
@@ -70,3 +71,51 @@ class Table:
 
     def default_option(self) -> AddOption:
         return AddOption(worker_id=self.zoo.worker_id)
+
+    # ---- streaming sharded checkpoint I/O ----
+    # Byte layout stays identical to the reference's per-shard Store
+    # (array_table.cpp:144-151 / matrix_table.cpp:457-464 write raw shard
+    # bytes in server order == the whole table row-major, since sharding
+    # is contiguous). Unlike round 1 there is NO rank-0 full-table
+    # staging: every rank pwrites its own shard at its byte offset in
+    # bounded chunks, so a table larger than one rank's free memory
+    # checkpoints fine.
+    _IO_CHUNK_BYTES = (64 << 20)  # per-chunk host staging bound
+
+    def _store_shard_stream(self, path: str, flat_shard: torch.Tensor,
+                            elem_offset: int, total_elems: int) -> None:
+        elem = flat_shard.element_size()
+        if self.zoo.rank == 0:
+            with open(path, "wb") as f:
+                f.truncate(total_elems * elem)
+        self.zoo.barrier()   # file exists at full size before any pwrite
+        chunk_elems = max(self._IO_CHUNK_BYTES // elem, 1)
+        n = flat_shard.numel()
+        if n:
+            with open(path, "r+b") as f:
+                f.seek(elem_offset * elem)
+                for off in range(0, n, chunk_elems):
+                    piece = flat_shard[off:off + chunk_elems]
+                    f.write(piece.cpu().numpy().tobytes())
+        self.zoo.barrier()
+
+    def _load_shard_stream(self, path: str, flat_shard: torch.Tensor,
+                           elem_offset: int, total_elems: int) -> None:
+        import numpy as np
+        elem = flat_shard.element_size()
+        CHECK(os.path.getsize(path) == total_elems * elem,
+              f"checkpoint size mismatch: {path} has "
+              f"{os.path.getsize(path)} bytes, table needs "
+              f"{total_elems * elem}")
+        chunk_elems = max(self._IO_CHUNK_BYTES // elem, 1)
+        n = flat_shard.numel()
+        np_dtype = np.dtype(str(flat_shard.dtype).replace("torch.", ""))
+        if n:
+            with open(path, "rb") as f:
+                f.seek(elem_offset * elem)
+                for off in range(0, n, chunk_elems):
+                    cnt = min(chunk_elems, n - off)
+                    buf = np.fromfile(f, dtype=np_dtype, count=cnt)
+                    flat_shard[off:off + cnt].copy_(
+                        torch.from_numpy(buf).to(flat_shard.device))
+        self.zoo.barrier()

@@ -281,18 +281,15 @@ class MatrixTable(Table):
 
     # ---- checkpoint (matrix_table.cpp:457-464) ----
     def store(self, path: str) -> None:
-        full = self.get()
-        if self.zoo.rank == 0:
-            full.cpu().numpy().tofile(path)
-        self.zoo.barrier()
+        """Raw whole-table bytes (row-major), streamed: each rank pwrites
+        its own row range at its byte offset — no rank-0 staging."""
+        self.flush()
+        self._store_shard_stream(path, self.shard.view(-1),
+                                 self.row_offset * self.num_col,
+                                 self.num_row * self.num_col)
 
     def load(self, path: str) -> None:
         self.flush()
-        import numpy as np
-        arr = np.fromfile(path, dtype=str(self.dtype).replace("torch.", ""))
-        CHECK(arr.size == self.num_row * self.num_col,
-              "checkpoint size mismatch")
-        full = torch.from_numpy(arr).view(self.num_row, self.num_col)
-        self.shard.copy_(full[self.row_offset:
-                              self.row_offset + self.local_rows].to(self.device))
-        self.zoo.barrier()
+        self._load_shard_stream(path, self.shard.view(-1),
+                                self.row_offset * self.num_col,
+                                self.num_row * self.num_col)
