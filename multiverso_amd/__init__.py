@@ -132,6 +132,11 @@ class ArrayTableHandler:
             if not is_master_worker():
                 init = torch.zeros_like(init)
             self.add(init, sync=True)
+            # async mode: the master's init add must be VISIBLE to every
+            # worker before any of them proceeds (in sync mode the add is
+            # already a collective; the barrier is then a cheap no-op
+            # beyond the reference's own post-init barrier)
+            barrier()
 
     def get(self) -> torch.Tensor:
         return self._table.get()
@@ -159,6 +164,7 @@ class MatrixTableHandler:
             if not is_master_worker():
                 init = torch.zeros_like(init)
             self.add(init, sync=True)
+            barrier()   # see ArrayTableHandler: init visible to all
 
     def get(self, row_ids: Optional[Sequence[int]] = None) -> torch.Tensor:
         if row_ids is None:

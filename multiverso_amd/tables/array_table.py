@@ -35,8 +35,12 @@ class ArrayTable(Table):
         self.size = size
         self.dtype = dtype
         self.spec = ShardSpec(size, self.zoo.num_servers)
-        off, cnt = self.spec.range_of(self.zoo.server_id)
+        if self.zoo.is_server:
+            off, cnt = self.spec.range_of(self.zoo.server_id)
+        else:
+            cnt = 0   # ps_role=worker: no shard hosted here
         self.shard = torch.zeros(cnt, dtype=dtype, device=self.device)
+        self.row_offset = 0   # keyed-op base (arrays have no keyed ops)
         self._make_updater(self.shard)
         # single-rank Add deferral + Add/Get fusion — same mechanism and
         # semantics as MatrixTable (see matrix_table.py)
@@ -61,6 +65,9 @@ class ArrayTable(Table):
     def get(self, out: Optional[torch.Tensor] = None,
             async_op: bool = False):
         """Whole-table Get (array_table.cpp:24-66 semantics)."""
+        eng = self.engine
+        if eng is not None:
+            return self._engine_get(eng, out, async_op)
         d = self._deferred
         if d is not None and not async_op:
             self._deferred = None
@@ -98,12 +105,48 @@ class ArrayTable(Table):
             return out, h
         return out
 
+    def _engine_get(self, eng, out, async_op):
+        """Async-mode Get: request each server's shard; served on
+        arrival (worker.cpp:30-51 -> server.cpp:36-46)."""
+        CHECK(self.zoo.is_worker, "ps_role=server ranks issue no Gets")
+        self.flush()
+        user_out = out
+        if out is None or not out.is_contiguous():
+            out = torch.empty(self.size, dtype=self.dtype,
+                              device=self.device)
+        CHECK(out.numel() == self.size, "Get buffer size mismatch")
+        with monitor("worker.get"):
+            pr = eng.whole_get(self, out.view(-1), 1)
+
+        def _finish() -> None:
+            if user_out is not None and user_out is not out:
+                user_out.copy_(out.view_as(user_out))
+        h = Handle(pr, _finish)
+        ret = user_out if user_out is not None else out
+        if async_op:
+            self._track(h)
+            return ret, h
+        h.wait()
+        return ret
+
     def add(self, delta: torch.Tensor, option: Optional[AddOption] = None,
             async_op: bool = False) -> Handle:
         """Whole-table Add: reduce-scatter + updater (server.cpp:48 →
-        array_table.cpp:116-127)."""
+        array_table.cpp:116-127); in async mode, per-server slices
+        applied on arrival."""
         CHECK(delta.numel() == self.size, "Add delta size mismatch")
         delta = delta.to(self.device, self.dtype).contiguous().view(-1)
+        eng = self.engine
+        if eng is not None:
+            CHECK(self.zoo.is_worker, "ps_role=server ranks issue no Adds")
+            self.flush()
+            with monitor("worker.add"):
+                h = Handle(eng.whole_add(self, delta, 1, option,
+                                         want_ack=not async_op))
+            if async_op:
+                return self._track(h)
+            h.wait()
+            return h
         if self.zoo.size == 1 and self.shard.is_cuda:
             self.flush()                 # at most one deferred Add
             self._deferred = (delta, option, delta._version)

@@ -18,6 +18,7 @@ observes every Add issued before it on this worker.
 from __future__ import annotations
 
 import os
+import threading
 from typing import List, Optional
 
 import torch
@@ -48,7 +49,34 @@ class Table:
         self.updater_type = updater_type
         self.updater = None  # created by subclass once the shard exists
         self._pending: List[Handle] = []
+        # serializes the async server thread against local worker ops on
+        # the same shard (the reference's one-consumer-thread-per-actor
+        # guarantee, actor.cpp:38-50, collapsed to a lock)
+        self._shard_lock = threading.Lock()
         _server_tables.append(self)
+
+    @property
+    def engine(self):
+        """The async PS engine, or None in sync/single-process mode."""
+        return self.zoo.async_engine
+
+    # ---- server-side entry points (ServerTable::ProcessAdd/ProcessGet,
+    # table_interface.h:61-75) — called by the async server thread AND by
+    # local worker fast paths; both hold the shard lock ----
+    def _server_apply_chunk(self, chunk: torch.Tensor, option) -> None:
+        with self._shard_lock:
+            self.updater.update(
+                chunk.to(self.updater.shard.device,
+                         self.updater.shard.dtype), option)
+
+    def _server_read_chunk_into(self, out_flat: torch.Tensor) -> None:
+        with self._shard_lock:
+            src = self.updater.shard.view(-1)
+            if out_flat.device == src.device:
+                out_flat.copy_(src)
+            else:
+                out_flat.copy_(src.cpu() if out_flat.device.type == "cpu"
+                               else src.to(out_flat.device))
 
     def _make_updater(self, shard: torch.Tensor) -> None:
         self.updater = create_updater(self.updater_type, shard)

@@ -32,12 +32,39 @@ class KVTable(Table):
     def _owner(self, key: int) -> int:
         return key % self.zoo.num_servers
 
+    @property
+    def num_shards(self) -> int:
+        return self.zoo.num_servers
+
     def raw(self) -> Dict[int, float]:
         return self._mirror
 
+    # ---- server-side entry points (kv_table.h:99-106) ----
+    def _server_kv_add(self, keys_t, vals_t) -> None:
+        with self._shard_lock:
+            for k, v in zip(keys_t.tolist(), vals_t.tolist()):
+                self._store[k] = self._store.get(k, 0) + v
+
+    def _server_kv_get(self, keys_t):
+        import torch
+        with self._shard_lock:
+            return torch.tensor([self._store.get(k, 0)
+                                 for k in keys_t.tolist()],
+                                dtype=torch.float64)
+
     def add(self, keys: Iterable[int], values: Iterable[float]) -> None:
-        """Collective: every rank contributes its (keys, values); each
-        server applies the adds for its keys."""
+        """Every rank contributes its (keys, values); each server applies
+        the adds for its keys. Async mode: p2p to the owners, served on
+        arrival (no coordination). Sync mode: collective."""
+        eng = self.engine
+        if eng is not None:
+            import torch
+            from ..comm import Handle
+            k = torch.as_tensor(list(keys), dtype=torch.int64)
+            v = torch.as_tensor(list(values), dtype=torch.float64)
+            if k.numel():
+                self._track(Handle(eng.kv_add(self, k, v)))
+            return
         mine: List = list(zip(keys, values))
         if dist.is_initialized() and self.zoo.size > 1:
             gathered: List = [None] * self.zoo.size
@@ -50,8 +77,18 @@ class KVTable(Table):
                     self._store[k] = self._store.get(k, 0) + v
 
     def get(self, keys: Iterable[int]) -> Dict[int, float]:
-        """Collective: pull requested keys into the local mirror."""
+        """Pull requested keys into the local mirror (async: p2p to the
+        owners; sync: collective)."""
         keys = list(keys)
+        eng = self.engine
+        if eng is not None:
+            import torch
+            self.flush()   # my own in-flight adds land first (FIFO)
+            k = torch.as_tensor(keys, dtype=torch.int64)
+            vals = eng.kv_get(self, k).tolist() if keys else []
+            for k_, v_ in zip(keys, vals):
+                self._mirror[k_] = v_
+            return {k_: self._mirror[k_] for k_ in keys}
         if dist.is_initialized() and self.zoo.size > 1:
             # each server broadcasts its shard's answers for all requests
             wanted: List = [None] * self.zoo.size
@@ -75,11 +112,15 @@ class KVTable(Table):
 
     # ---- checkpoint ----
     def store(self, path: str) -> None:
+        self.flush()
+        self.zoo.barrier()   # async mode: every worker's adds are applied
+        with self._shard_lock:
+            snapshot = dict(self._store)
         if dist.is_initialized() and self.zoo.size > 1:
             shards: List = [None] * self.zoo.size
-            dist.all_gather_object(shards, self._store)
+            dist.all_gather_object(shards, snapshot)
         else:
-            shards = [self._store]
+            shards = [snapshot]
         if self.zoo.rank == 0:
             merged: Dict[int, float] = {}
             for s in shards:

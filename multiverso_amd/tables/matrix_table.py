@@ -45,7 +45,11 @@ class MatrixTable(Table):
         self.num_col = num_col
         self.dtype = dtype
         self.spec = ShardSpec(num_row, self.zoo.num_servers)
-        self.row_offset, self.local_rows = self.spec.range_of(self.zoo.server_id)
+        if self.zoo.is_server:
+            self.row_offset, self.local_rows = self.spec.range_of(
+                self.zoo.server_id)
+        else:
+            self.row_offset, self.local_rows = 0, 0  # ps_role=worker
         self.shard = torch.zeros(self.local_rows, num_col, dtype=dtype,
                                  device=self.device)
         if random_init is not None:
@@ -81,7 +85,34 @@ class MatrixTable(Table):
         super().flush()
 
     # ---- whole-table ops ----
+    def _engine_get(self, eng, out, async_op):
+        """Async-mode whole-table Get (worker.cpp:30-51 semantics)."""
+        CHECK(self.zoo.is_worker, "ps_role=server ranks issue no Gets")
+        self.flush()
+        user_out = out
+        if out is None or not out.is_contiguous():
+            out = torch.empty(self.num_row, self.num_col, dtype=self.dtype,
+                              device=self.device)
+        CHECK(out.numel() == self.num_row * self.num_col,
+              "Get buffer size mismatch")
+        with monitor("worker.get"):
+            pr = eng.whole_get(self, out.view(-1), self.num_col)
+
+        def _finish() -> None:
+            if user_out is not None and user_out is not out:
+                user_out.copy_(out.view_as(user_out))
+        h = Handle(pr, _finish)
+        ret = user_out if user_out is not None else out
+        if async_op:
+            self._track(h)
+            return ret, h
+        h.wait()
+        return ret
+
     def get(self, out: Optional[torch.Tensor] = None, async_op: bool = False):
+        eng = self.engine
+        if eng is not None:
+            return self._engine_get(eng, out, async_op)
         d = self._deferred
         if d is not None and not async_op:
             self._deferred = None
@@ -136,6 +167,17 @@ class MatrixTable(Table):
         CHECK(delta.numel() == self.num_row * self.num_col,
               "Add delta size mismatch")
         delta = delta.to(self.device, self.dtype).contiguous().view(-1)
+        eng = self.engine
+        if eng is not None:
+            CHECK(self.zoo.is_worker, "ps_role=server ranks issue no Adds")
+            self.flush()
+            with monitor("worker.add"):
+                h = Handle(eng.whole_add(self, delta, self.num_col, option,
+                                         want_ack=not async_op))
+            if async_op:
+                return self._track(h)
+            h.wait()
+            return h
         if self.zoo.size == 1 and self.shard.is_cuda:
             self.flush()                 # at most one deferred Add
             self._deferred = (delta, option, delta._version)
@@ -234,6 +276,12 @@ class MatrixTable(Table):
         ids cost one D2H) and the split sizes ride the gloo control lane —
         no device sync between launch and the value all-to-all."""
         self.flush()
+        eng = self.engine
+        if eng is not None:
+            CHECK(self.zoo.is_worker, "ps_role=server ranks issue no Gets")
+            ids = torch.as_tensor(row_ids, dtype=torch.int64).cpu()
+            with monitor("worker.get_rows"):
+                return eng.keyed_get(self, ids, self.num_col)
         ids = torch.as_tensor(row_ids, dtype=torch.int64)
         with monitor("worker.get_rows"):
             in_ids, _, recv_sizes, order, send_sizes = all_to_all_rows(
@@ -268,6 +316,13 @@ class MatrixTable(Table):
         vals = values.to(self.device, self.dtype).contiguous()
         CHECK(vals.numel() == ids.numel() * self.num_col,
               "add_rows values size mismatch")
+        eng = self.engine
+        if eng is not None:
+            CHECK(self.zoo.is_worker, "ps_role=server ranks issue no Adds")
+            with monitor("worker.add_rows"):
+                self._track(Handle(eng.keyed_add(
+                    self, ids.cpu(), vals, self.num_col, option)))
+            return
         with monitor("worker.add_rows"):
             in_ids, in_vals, _, _, _ = all_to_all_rows(
                 ids, vals.view(-1), self.spec, self.num_col,
@@ -278,6 +333,17 @@ class MatrixTable(Table):
                     self._scatter_update_local(
                         local, in_vals.view(-1, self.num_col), option,
                         assume_unique=assume_unique and self.zoo.size == 1)
+
+    # ---- server-side keyed entry points (async engine + local path) ----
+    def _server_add_rows(self, local_ids: torch.Tensor,
+                         vals2d: torch.Tensor, option) -> None:
+        with self._shard_lock:
+            with monitor("server.update_rows"):
+                self._scatter_update_local(local_ids, vals2d, option)
+
+    def _server_get_rows(self, local_ids: torch.Tensor) -> torch.Tensor:
+        with self._shard_lock:
+            return self._gather_local(local_ids)
 
     # ---- checkpoint (matrix_table.cpp:457-464) ----
     def store(self, path: str) -> None:

@@ -35,6 +35,14 @@ class SparseMatrixTable(MatrixTable):
                  updater_type: Optional[str] = None,
                  random_init=None) -> None:
         super().__init__(num_row, num_col, dtype, updater_type, random_init)
+        # The stale-aware whole-table Get is a coordinated protocol
+        # (every owner builds every requester's stale list in lockstep,
+        # sparse_matrix_table.cpp:226-258); it runs on the BSP collective
+        # plane. Async deployments use MatrixTable row ops instead.
+        CHECK(self.engine is None,
+              "SparseMatrixTable requires sync mode (-sync=true): its "
+              "stale-row exchange is collective; async-mode training "
+              "should use MatrixTable.get_rows/add_rows")
         # SparseFilter on the stale-row reply payloads (the reference
         # filters every outgoing sparse-table bundle,
         # sparse_matrix_table.cpp:148-153): per-destination segments are

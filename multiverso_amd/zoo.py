@@ -63,6 +63,14 @@ class Zoo:
         # reference's ZMQ Bind/Connect replaced mpirun
         self._bound_rank: Optional[int] = None
         self._endpoints: Optional[List[str]] = None
+        # async PS engine (the reference's DEFAULT mode, server.cpp:36-58):
+        # active when sync=false and world>1 — table ops are served on
+        # arrival by per-rank server threads instead of collectives
+        self.async_engine = None
+        # ps_role bookkeeping (zoo.cpp:23,29-35): which global ranks host
+        # shards / issue worker ops; role ALL (default) -> all of them
+        self.server_ranks: List[int] = [0]
+        self.worker_ranks: List[int] = [0]
 
     # ---- explicit rendezvous (MV_NetBind / MV_NetConnect) ----
     def net_bind(self, rank: int, endpoint: str) -> bool:
@@ -102,9 +110,8 @@ class Zoo:
         role = get_flag("ps_role")
         self.role = {"default": Role.ALL, "worker": Role.WORKER,
                      "server": Role.SERVER, "none": Role.NONE}.get(role, Role.ALL)
-        CHECK(self.role == Role.ALL,
-              "MI355X rebuild runs every rank as worker+server (role=default); "
-              f"got ps_role={role}")
+        CHECK(self.role in (Role.ALL, Role.WORKER, Role.SERVER),
+              f"unsupported ps_role={role}")
 
         explicit = (self._bound_rank is not None
                     and self._endpoints is not None)
@@ -164,6 +171,37 @@ class Zoo:
         else:
             self.rank, self.size = 0, 1
 
+        # role map (zoo.cpp:29-35): exchange every rank's role over the
+        # host lane, derive the worker/server rank lists that shard specs
+        # and the async engine address
+        if self.size > 1:
+            vec = torch.zeros(self.size, dtype=torch.int64)
+            vec[self.rank] = self.role
+            dist.all_reduce(vec, group=self.control_pg)
+            roles = vec.tolist()
+        else:
+            roles = [self.role]
+        self.server_ranks = [r for r, ro in enumerate(roles)
+                             if ro & Role.SERVER]
+        self.worker_ranks = [r for r, ro in enumerate(roles)
+                             if ro & Role.WORKER]
+        CHECK(len(self.server_ranks) > 0,
+              "at least one rank must host shards (ps_role=server|default)")
+        CHECK(len(self.worker_ranks) > 0,
+              "at least one rank must be a worker (ps_role=worker|default)")
+        split = (len(self.server_ranks) != self.size
+                 or len(self.worker_ranks) != self.size)
+        if self.size > 1 and not bool(get_flag("sync")):
+            # the reference's default (async) mode: per-rank server
+            # threads serve Get/Add on arrival; -sync=true selects the
+            # BSP collective plane instead
+            from .async_ps import AsyncEngine
+            self.async_engine = AsyncEngine(self)
+        CHECK(not split or self.async_engine is not None,
+              "ps_role worker/server splits require the async PS mode "
+              "(sync=false): the BSP collective plane needs every rank "
+              "to be worker+server")
+
         lvl = str(get_flag("log_level")).lower()
         from .log import LogLevel
         if lvl in ("debug", "info", "error", "fatal"):
@@ -184,7 +222,15 @@ class Zoo:
         """MV_ShutDown equivalent (zoo.cpp:104-161)."""
         if not self.started:
             return
-        self.barrier()
+        if self.async_engine is not None:
+            eng = self.async_engine
+            eng.shutdown()     # drain + barrier + FinishTrain + join
+            self.async_engine = None
+            if dist.is_initialized():
+                dist.destroy_process_group(eng.req)
+                dist.destroy_process_group(eng.rep)
+        else:
+            self.barrier()
         from .tables.base import free_tables
         free_tables()
         self._tables.clear()
@@ -199,19 +245,33 @@ class Zoo:
     # ---- bookkeeping (zoo.h:19-85) ----
     @property
     def num_workers(self) -> int:
-        return self.size
+        return len(self.worker_ranks)
 
     @property
     def num_servers(self) -> int:
-        return self.size
+        return len(self.server_ranks)
 
     @property
     def worker_id(self) -> int:
-        return self.rank
+        try:
+            return self.worker_ranks.index(self.rank)
+        except ValueError:
+            return -1
 
     @property
     def server_id(self) -> int:
-        return self.rank
+        try:
+            return self.server_ranks.index(self.rank)
+        except ValueError:
+            return -1
+
+    @property
+    def is_worker(self) -> bool:
+        return self.rank in self.worker_ranks
+
+    @property
+    def is_server(self) -> bool:
+        return self.rank in self.server_ranks
 
     def register_table(self, table) -> int:
         tid = self._next_table_id
@@ -221,6 +281,10 @@ class Zoo:
 
     # ---- collectives ----
     def barrier(self) -> None:
+        if self.async_engine is not None:
+            # reference MV_Barrier meaning preserved under async: all
+            # work issued before the barrier is visible after it
+            self.async_engine.drain()
         if dist.is_initialized():
             if self.backend == "nccl":
                 dist.barrier(device_ids=[self.device.index])

@@ -1,0 +1,207 @@
+"""True-async PS mode (sync=False, the reference's default — SURVEY.md
+§2.8 "Async data parallelism"): table ops are served on ARRIVAL by
+per-rank server threads (multiverso_amd/async_ps.py); workers never wait
+for each other. These tests pin the VERDICT r1 "done" criteria:
+
+- sync=False observably differs from sync=True (independence test),
+- ranks issuing DIFFERENT op sequences complete without deadlock and
+  with reference-consistent values after a drain barrier,
+- ps_role worker/server splits work (reference zoo.cpp:23,29-35).
+"""
+
+import time
+
+import numpy as np
+import pytest
+import torch
+
+from conftest import run_dist
+
+
+def _unequal_rounds(rank, world):
+    """Rank 0 issues 3 Add+Get rounds while rank 1 issues 1 — the exact
+    scenario that deadlocks a collective plane. After the shutdown
+    drain, a fresh session sees every add exactly once."""
+    import multiverso_amd as mv
+    mv.init()
+    t = mv.ArrayTable(10)
+    rounds = 3 if rank == 0 else 1
+    for _ in range(rounds):
+        t.add(torch.full((10,), 1.0))
+        got = t.get()
+        # FIFO per (worker, server) pair: my own adds are always visible
+        assert float(got.min()) >= 1.0
+    mv.barrier()
+    got = t.get()
+    expect = 3.0 + (world - 1) * 1.0  # rank0's 3 + one from each other
+    assert torch.equal(got, torch.full((10,), expect)), (rank, got)
+    mv.shutdown()
+
+
+def test_async_unequal_rounds():
+    run_dist(_unequal_rounds, 2)
+
+
+def test_async_unequal_rounds_ws3():
+    run_dist(_unequal_rounds, 3)
+
+
+def _independence(rank, world):
+    """sync=False MUST differ from sync=True: rank 0 completes many
+    Get/Add rounds while rank 1 is asleep. Under the BSP collective
+    plane this would block until rank 1 joined (~2s); async must finish
+    far sooner."""
+    import multiverso_amd as mv
+    mv.init()
+    t = mv.ArrayTable(64)
+    if rank == 1:
+        time.sleep(2.0)
+        t.add(torch.ones(64))
+    else:
+        t0 = time.perf_counter()
+        for _ in range(20):
+            t.add(torch.ones(64))
+            t.get()
+        elapsed = time.perf_counter() - t0
+        assert elapsed < 1.0, f"async rounds blocked on straggler: {elapsed}s"
+    mv.barrier()
+    got = t.get()
+    assert torch.equal(got, torch.full((64,), 20.0 + (world - 1)))
+    mv.shutdown()
+
+
+def test_async_independence():
+    run_dist(_independence, 2)
+
+
+def _keyed_rows(rank, world):
+    """Row-keyed Get/Add served on arrival; different ranks touch
+    different (and overlapping) rows with different call counts."""
+    import multiverso_amd as mv
+    mv.init()
+    t = mv.MatrixTable(9, 4)   # uneven shards at ws 2: 4 + 5 rows
+    if rank == 0:
+        t.add_rows([0, 8], torch.ones(2, 4))
+        t.add_rows([0], torch.ones(1, 4))
+    else:
+        t.add_rows([8], torch.full((1, 4), 2.0))
+    got = t.get_rows([0, 8] if rank == 0 else [8])
+    assert got.shape[1] == 4
+    mv.barrier()
+    got = t.get_rows([0, 4, 8])
+    expect = torch.zeros(3, 4)
+    expect[0] = 2.0                      # rank 0 added twice
+    expect[2] = 1.0 + 2.0 * (world - 1)  # both ranks touched row 8
+    assert torch.equal(got, expect), (rank, got)
+    mv.shutdown()
+
+
+def test_async_keyed_rows():
+    run_dist(_keyed_rows, 2)
+
+
+def _kv_async(rank, world):
+    import multiverso_amd as mv
+    mv.init()
+    t = mv.KVTable()
+    t.add([1, 2, 3 + rank], [1.0, 2.0, 1.0])
+    got = t.get([1])   # FIFO: my own add visible
+    assert got[1] >= 1.0
+    mv.barrier()
+    got = t.get([1, 2, 3, 4])
+    assert got[1] == 1.0 * world and got[2] == 2.0 * world
+    mv.shutdown()
+
+
+def test_async_kv():
+    run_dist(_kv_async, 2)
+
+
+def _whole_matrix_async(rank, world):
+    """Whole-table ops with per-server slicing + momentum updater state
+    living where the shard lives."""
+    import multiverso_amd as mv
+    from multiverso_amd import AddOption
+    mv.init()
+    t = mv.MatrixTable(7, 3, updater_type="default")   # shards 3 + 4
+    delta = torch.arange(21, dtype=torch.float32).view(7, 3)
+    t.add(delta)          # sync add: ack means applied
+    got = t.get()
+    assert torch.all(got >= delta), (rank, got)  # mine applied, maybe more
+    mv.barrier()
+    got = t.get()
+    assert torch.equal(got, delta * world), (rank, got)
+    mv.shutdown()
+
+
+def test_async_whole_matrix():
+    run_dist(_whole_matrix_async, 2)
+
+
+def _role_split(rank, world):
+    """ps_role deployment split (zoo.cpp:23,29-35): rank 2 is a pure
+    server (hosts ALL shards, issues no ops); ranks 0-1 are pure workers
+    (host nothing). The reference's dedicated-server topology."""
+    import multiverso_amd as mv
+    role = "server" if rank == 2 else "worker"
+    mv.init([f"-ps_role={role}"])
+    assert mv.servers_num() == 1 and mv.workers_num() == 2
+    t = mv.ArrayTable(10)
+    if role == "worker":
+        assert t.shard.numel() == 0      # nothing hosted on workers
+        t.add(torch.full((10,), 2.0))
+        got = t.get()
+        assert float(got.min()) >= 2.0   # my add visible
+    else:
+        assert t.shard.numel() == 10     # the whole table lives here
+    mv.barrier()
+    if role == "worker":
+        got = t.get()
+        assert torch.equal(got, torch.full((10,), 2.0 * 2)), (rank, got)
+    mv.shutdown()
+
+
+def test_ps_role_split():
+    run_dist(_role_split, 3)
+
+
+def _dcasgd_per_worker_slots(rank, world):
+    """DC-ASGD's per-worker weight backups (AddOption.worker_id selects
+    the slot) — genuinely exercised only in async mode, where each
+    worker's add arrives with its own id (the reference's intended
+    operation; updater.cpp:51-54)."""
+    import multiverso_amd as mv
+    from multiverso_amd import AddOption
+    mv.init()
+    t = mv.ArrayTable(6, updater_type="dcasgd")
+    opt = AddOption(worker_id=mv.worker_id(), learning_rate=0.1,
+                    lambda_=0.0)   # lambda=0 -> plain SGD, exact oracle
+    for _ in range(2):
+        t.add(torch.ones(6), option=opt)
+    mv.barrier()
+    got = t.get()
+    expect = torch.full((6,), -0.1 * 2 * world)
+    assert torch.allclose(got, expect), (rank, got)
+    # each worker's backup slot exists on every server shard it touched
+    assert len(t.updater._bak) >= 1 or t.shard.numel() == 0
+    mv.shutdown()
+
+
+def test_async_dcasgd_slots():
+    run_dist(_dcasgd_per_worker_slots, 2)
+
+
+def _handler_async(rank, world):
+    """Binding handler init protocol under async mode: master adds the
+    init value, others add zeros; the post-init barrier makes it visible
+    exactly once to every worker before any proceeds."""
+    import multiverso_amd as mv
+    mv.init()
+    h = mv.ArrayTableHandler(8, init_value=np.full(8, 3.0, dtype=np.float32))
+    got = h.get()
+    assert torch.equal(got, torch.full((8,), 3.0)), (rank, got)
+    mv.shutdown()
+
+
+def test_async_handler_protocol():
+    run_dist(_handler_async, 2)

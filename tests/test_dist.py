@@ -99,12 +99,18 @@ def _handler_init_protocol(rank, world):
 
 
 def _async_pipeline(rank, world):
+    # default (sync=False) is the TRUE ASYNC mode at world>1: my get is
+    # only guaranteed to see MY prior adds; a barrier (drain) makes
+    # everyone's adds visible (reference MV_Barrier semantics)
     import multiverso_amd as mv
     mv.init()
     t = mv.ArrayTable(16)
     for _ in range(4):
         t.add(torch.ones(16), async_op=True)
-    got = t.get()  # flushes pending
+    got = t.get()  # FIFO per server: sees at least my own 4 adds
+    assert float(got.min()) >= 4.0, got
+    mv.barrier()
+    got = t.get()
     assert torch.equal(got, torch.full((16,), 4.0 * world))
     mv.shutdown()
 

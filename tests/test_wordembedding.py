@@ -212,6 +212,10 @@ def _w2v_uneven_blocks(rank, world):
             break
         model.train_block(*(blk if blk is not None else empty))
         model.sync_word_count()
+    # default mode is now TRUE ASYNC: the global count is only exact
+    # after a drain barrier (mid-training reads may lag — by design)
+    mv.barrier()
+    model.sync_word_count()
     assert model.word_count_actual == 120  # 2*40 + 1*40
     mv.shutdown()
 
