@@ -135,8 +135,10 @@ class PSModel:
         local = pulled.clone()
         total_loss = 0.0
         adagrad = self.table.updater_type == "adagrad"
-        if self._fused_sigmoid(local):
-            total_loss = self._train_chunk_fused(batches, union, local)
+        fused = self._fused_kind(local)
+        if fused:
+            total_loss = self._train_chunk_fused(batches, union, local,
+                                                 fused)
         else:
             for b in batches:
                 lidx = torch.searchsorted(union, b.keys)
@@ -164,18 +166,30 @@ class PSModel:
             self.table.add_rows(union, pulled - local, assume_unique=True)
         return total_loss / max(len(batches), 1)
 
-    def _fused_sigmoid(self, local: torch.Tensor) -> bool:
-        """Fused K13/K14 minibatch kernels: sigmoid objective, single
-        output column, GPU. Other objectives use the torch path."""
-        from .objective import SigmoidObjective
-        return (local.is_cuda and self.cols == 1
-                and type(self.objective) is SigmoidObjective)
+    def _fused_kind(self, local: torch.Tensor) -> str:
+        """Which fused K13/K14 minibatch kernel pair serves this
+        objective on GPU ('' = torch path). All three reference
+        objectives now have fused forms: sigmoid (1 col), softmax
+        (2..64 classes), ftrl (z|n state, <=32 outputs)."""
+        from .objective import (FTRLObjective, SigmoidObjective,
+                                SoftmaxObjective)
+        if not local.is_cuda:
+            return ""
+        if self.cols == 1 and type(self.objective) is SigmoidObjective:
+            return "sigmoid"
+        if (type(self.objective) is SoftmaxObjective
+                and 2 <= self.cols <= 64):
+            return "softmax"
+        if (type(self.objective) is FTRLObjective
+                and self.cfg.output_size <= 32):
+            return "ftrl"
+        return ""
 
-    def _train_chunk_fused(self, batches, union, local) -> float:
-        """Two HIP kernels per minibatch (k_lr_sigmoid_fwd/_scatter)
-        instead of ~20 torch ops; numerics match objective.gradient +
-        scatter (tests/test_gpu_kernels.py). Returns summed per-batch
-        mean losses (one host sync per chunk instead of per minibatch)."""
+    def _train_chunk_fused(self, batches, union, local, kind) -> float:
+        """Two HIP kernels per minibatch (fwd + scatter) instead of ~20
+        torch ops; numerics match objective.gradient + scatter
+        (tests/test_gpu_kernels.py). Returns summed per-batch mean
+        losses (one host sync per chunk instead of per minibatch)."""
         from multiverso_amd import ops as _ops
         hip = _ops.module(required=True)
         from .objective import L1Regular, L2Regular
@@ -185,20 +199,39 @@ class PSModel:
             reg_type, reg_coef = 1, reg.coef
         elif isinstance(reg, L2Regular):
             reg_type, reg_coef = 2, reg.coef
+        K = self.cfg.output_size
         wflat = local.view(-1)
         loss_acc = torch.zeros((), device=self.device)
         for b in batches:
             lidx = torch.searchsorted(union, b.keys)
             ptr32 = b.ptr.to(torch.int32)
             B = b.size
-            err = torch.empty(B, device=self.device)
             lossb = torch.empty(B, device=self.device)
             wts = None if b.weights is None else b.weights.float()
-            hip.lr_sigmoid_forward(wflat, lidx, b.vals, ptr32,
-                                   b.labels.float(), wts, err, lossb)
-            lr = self.sched.next_lr()
-            hip.lr_sigmoid_scatter(wflat, lidx, b.vals, ptr32, err, lr,
-                                   reg_type, reg_coef)
+            labels = b.labels.float()
+            if kind == "sigmoid":
+                err = torch.empty(B, device=self.device)
+                hip.lr_sigmoid_forward(wflat, lidx, b.vals, ptr32, labels,
+                                       wts, err, lossb)
+                lr = self.sched.next_lr()
+                hip.lr_sigmoid_scatter(wflat, lidx, b.vals, ptr32, err, lr,
+                                       reg_type, reg_coef)
+            elif kind == "softmax":
+                err = torch.empty(B * K, device=self.device)
+                hip.lr_softmax_forward(wflat, lidx, b.vals, ptr32, labels,
+                                       wts, err, lossb, K)
+                lr = self.sched.next_lr()
+                hip.lr_softmax_scatter(wflat, lidx, b.vals, ptr32, err, lr,
+                                       reg_type, reg_coef, K)
+            else:  # ftrl: state update is lr-free (updater.cpp:79-101)
+                o = self.objective
+                err = torch.empty(B * K, device=self.device)
+                hip.lr_ftrl_forward(wflat, lidx, b.vals, ptr32, labels, wts,
+                                    err, lossb, 1.0 / o.alpha_inv, o.beta,
+                                    o.l1, o.l2, K)
+                hip.lr_ftrl_scatter(wflat, lidx, b.vals, ptr32, err,
+                                    1.0 / o.alpha_inv, o.beta, o.l1, o.l2,
+                                    K)
             loss_acc += lossb.mean()
         return float(loss_acc)
 

@@ -51,6 +51,29 @@ void mv_launch_lr_sigmoid_scatter(float*, const int64_t*, const float*,
 void mv_launch_row_scatter_adagrad(float*, float*, const float*,
                                    const int64_t*, float, float, float,
                                    int64_t, int64_t, int, hipStream_t);
+void mv_launch_lr_softmax_fwd(const float*, const int64_t*, const float*,
+                              const int*, const float*, const float*,
+                              float*, float*, int64_t, int64_t, hipStream_t);
+void mv_launch_lr_softmax_scatter(float*, const int64_t*, const float*,
+                                  const int*, const float*, float, int,
+                                  float, int64_t, int64_t, hipStream_t);
+void mv_launch_lr_ftrl_fwd(const float*, const int64_t*, const float*,
+                           const int*, const float*, const float*, float*,
+                           float*, float, float, float, float, int64_t,
+                           int64_t, hipStream_t);
+void mv_launch_lr_ftrl_scatter(float*, const int64_t*, const float*,
+                               const int*, const float*, float, float,
+                               float, float, int64_t, int64_t, hipStream_t);
+void mv_launch_add_f64(double*, const double*, int64_t, hipStream_t);
+void mv_launch_sgd_f64(double*, const double*, int64_t, hipStream_t);
+void mv_launch_momentum_f64(double*, double*, const double*, double, int64_t,
+                            hipStream_t);
+void mv_launch_adagrad_f64(double*, double*, const double*, double, double,
+                           double, int64_t, hipStream_t);
+void mv_launch_sgd_copy_f64(double*, const double*, double*, double, int64_t,
+                            hipStream_t);
+void mv_launch_add_i32(int32_t*, const int32_t*, int64_t, hipStream_t);
+void mv_launch_add_i64(int64_t*, const int64_t*, int64_t, hipStream_t);
 }
 
 namespace {
@@ -65,6 +88,13 @@ void check_f32(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
 }
 
+void check_dev(const torch::Tensor& t, const char* name,
+               torch::ScalarType st) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == st, name, " dtype mismatch");
+}
+
 void nt_copy(torch::Tensor dst, torch::Tensor src) {
   check_f32(dst, "dst"); check_f32(src, "src");
   TORCH_CHECK(dst.numel() == src.numel(), "size mismatch");
@@ -73,47 +103,97 @@ void nt_copy(torch::Tensor dst, torch::Tensor src) {
 }
 
 void add_inplace(torch::Tensor data, torch::Tensor delta) {
-  check_f32(data, "data"); check_f32(delta, "delta");
+  // dtype-dispatched K1 (reference instantiates int/float/double tables,
+  // array_table.cpp:153-154; int updater is add-only, updater.cpp:40-43)
+  auto st = data.scalar_type();
+  check_dev(data, "data", st); check_dev(delta, "delta", st);
   TORCH_CHECK(data.numel() == delta.numel(), "size mismatch");
-  mv_launch_add(data.data_ptr<float>(), delta.data_ptr<float>(),
-                data.numel(), cur_stream());
+  int64_t n = data.numel();
+  if (st == torch::kFloat32)
+    mv_launch_add(data.data_ptr<float>(), delta.data_ptr<float>(), n,
+                  cur_stream());
+  else if (st == torch::kFloat64)
+    mv_launch_add_f64(data.data_ptr<double>(), delta.data_ptr<double>(), n,
+                      cur_stream());
+  else if (st == torch::kInt32)
+    mv_launch_add_i32(data.data_ptr<int32_t>(), delta.data_ptr<int32_t>(), n,
+                      cur_stream());
+  else if (st == torch::kInt64)
+    mv_launch_add_i64(data.data_ptr<int64_t>(), delta.data_ptr<int64_t>(), n,
+                      cur_stream());
+  else
+    TORCH_CHECK(false, "add_inplace: unsupported dtype ", data.dtype());
 }
 
 void sgd_update(torch::Tensor data, torch::Tensor delta) {
-  check_f32(data, "data"); check_f32(delta, "delta");
+  auto st = data.scalar_type();
+  check_dev(data, "data", st); check_dev(delta, "delta", st);
   TORCH_CHECK(data.numel() == delta.numel(), "size mismatch");
-  mv_launch_sgd(data.data_ptr<float>(), delta.data_ptr<float>(),
-                data.numel(), cur_stream());
+  if (st == torch::kFloat32)
+    mv_launch_sgd(data.data_ptr<float>(), delta.data_ptr<float>(),
+                  data.numel(), cur_stream());
+  else if (st == torch::kFloat64)
+    mv_launch_sgd_f64(data.data_ptr<double>(), delta.data_ptr<double>(),
+                      data.numel(), cur_stream());
+  else
+    TORCH_CHECK(false, "sgd_update: unsupported dtype ", data.dtype());
 }
 
 void momentum_update(torch::Tensor data, torch::Tensor m, torch::Tensor delta,
                      double mu) {
-  check_f32(data, "data"); check_f32(m, "m"); check_f32(delta, "delta");
+  auto st = data.scalar_type();
+  check_dev(data, "data", st); check_dev(m, "m", st);
+  check_dev(delta, "delta", st);
   TORCH_CHECK(data.numel() == delta.numel() && data.numel() == m.numel(),
               "size mismatch");
-  mv_launch_momentum(data.data_ptr<float>(), m.data_ptr<float>(),
-                     delta.data_ptr<float>(), (float)mu, data.numel(),
-                     cur_stream());
+  if (st == torch::kFloat32)
+    mv_launch_momentum(data.data_ptr<float>(), m.data_ptr<float>(),
+                       delta.data_ptr<float>(), (float)mu, data.numel(),
+                       cur_stream());
+  else if (st == torch::kFloat64)
+    mv_launch_momentum_f64(data.data_ptr<double>(), m.data_ptr<double>(),
+                           delta.data_ptr<double>(), mu, data.numel(),
+                           cur_stream());
+  else
+    TORCH_CHECK(false, "momentum_update: unsupported dtype ", data.dtype());
 }
 
 void adagrad_update(torch::Tensor data, torch::Tensor gsq, torch::Tensor delta,
                     double lr, double rho, double eps) {
-  check_f32(data, "data"); check_f32(gsq, "gsq"); check_f32(delta, "delta");
+  auto st = data.scalar_type();
+  check_dev(data, "data", st); check_dev(gsq, "gsq", st);
+  check_dev(delta, "delta", st);
   TORCH_CHECK(data.numel() == delta.numel() && data.numel() == gsq.numel(),
               "size mismatch");
-  mv_launch_adagrad(data.data_ptr<float>(), gsq.data_ptr<float>(),
-                    delta.data_ptr<float>(), (float)lr, (float)rho, (float)eps,
-                    data.numel(), cur_stream());
+  if (st == torch::kFloat32)
+    mv_launch_adagrad(data.data_ptr<float>(), gsq.data_ptr<float>(),
+                      delta.data_ptr<float>(), (float)lr, (float)rho,
+                      (float)eps, data.numel(), cur_stream());
+  else if (st == torch::kFloat64)
+    mv_launch_adagrad_f64(data.data_ptr<double>(), gsq.data_ptr<double>(),
+                          delta.data_ptr<double>(), lr, rho, eps,
+                          data.numel(), cur_stream());
+  else
+    TORCH_CHECK(false, "adagrad_update: unsupported dtype ", data.dtype());
 }
 
 void sgd_copy_update(torch::Tensor data, torch::Tensor delta,
                      torch::Tensor out, double sign) {
-  check_f32(data, "data"); check_f32(delta, "delta"); check_f32(out, "out");
+  auto st = data.scalar_type();
+  check_dev(data, "data", st); check_dev(delta, "delta", st);
+  check_dev(out, "out", st);
   TORCH_CHECK(data.numel() == delta.numel() && data.numel() == out.numel(),
               "size mismatch");
-  mv_launch_sgd_copy(data.data_ptr<float>(), delta.data_ptr<float>(),
-                     out.data_ptr<float>(), (float)sign, data.numel(),
-                     cur_stream());
+  if (st == torch::kFloat32)
+    mv_launch_sgd_copy(data.data_ptr<float>(), delta.data_ptr<float>(),
+                       out.data_ptr<float>(), (float)sign, data.numel(),
+                       cur_stream());
+  else if (st == torch::kFloat64)
+    mv_launch_sgd_copy_f64(data.data_ptr<double>(), delta.data_ptr<double>(),
+                           out.data_ptr<double>(), sign, data.numel(),
+                           cur_stream());
+  else
+    TORCH_CHECK(false, "sgd_copy_update: unsupported dtype ", data.dtype());
 }
 
 void momentum_copy_update(torch::Tensor data, torch::Tensor m,
@@ -253,7 +333,7 @@ void w2v_train(torch::Tensor in_emb, torch::Tensor out_emb,
   check_f32(out_label, "out_label");
   TORCH_CHECK(in_emb.dim() == 2 && out_emb.dim() == 2, "emb must be 2-D");
   TORCH_CHECK(in_emb.size(1) == out_emb.size(1), "dim mismatch");
-  TORCH_CHECK(in_emb.size(1) <= 512, "w2v kernel supports dim <= 512");
+  TORCH_CHECK(in_emb.size(1) <= 2048, "w2v kernel supports dim <= 2048");
   TORCH_CHECK(in_idx.scalar_type() == torch::kInt64 &&
               out_idx.scalar_type() == torch::kInt64, "idx must be int64");
   TORCH_CHECK(in_off.scalar_type() == torch::kInt32 &&
@@ -285,7 +365,7 @@ void w2v_train_ns(torch::Tensor in_emb, torch::Tensor out_emb,
   check_f32(in_emb, "in_emb"); check_f32(out_emb, "out_emb");
   TORCH_CHECK(in_emb.dim() == 2 && out_emb.dim() == 2, "emb must be 2-D");
   TORCH_CHECK(in_emb.size(1) == out_emb.size(1), "dim mismatch");
-  TORCH_CHECK(in_emb.size(1) <= 512, "w2v kernel supports dim <= 512");
+  TORCH_CHECK(in_emb.size(1) <= 2048, "w2v kernel supports dim <= 2048");
   TORCH_CHECK(in_idx.scalar_type() == torch::kInt64 &&
               centers.scalar_type() == torch::kInt64 &&
               pool.scalar_type() == torch::kInt64, "ids must be int64");
@@ -361,6 +441,93 @@ void lr_sigmoid_scatter(torch::Tensor w, torch::Tensor keys,
                                cur_stream());
 }
 
+void lr_softmax_forward(torch::Tensor w, torch::Tensor keys,
+                        torch::Tensor vals, torch::Tensor ptr,
+                        torch::Tensor labels,
+                        c10::optional<torch::Tensor> wts,
+                        torch::Tensor err, torch::Tensor loss, int64_t K) {
+  check_f32(w, "w"); check_f32(vals, "vals"); check_f32(labels, "labels");
+  check_f32(err, "err"); check_f32(loss, "loss");
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64, "keys must be int64");
+  TORCH_CHECK(ptr.scalar_type() == torch::kInt32, "ptr must be int32");
+  TORCH_CHECK(K >= 2 && K <= 64,
+              "fused softmax supports 2..64 classes (got ", K, ")");
+  int64_t B = ptr.numel() - 1;
+  TORCH_CHECK(labels.numel() == B && loss.numel() == B, "batch mismatch");
+  TORCH_CHECK(err.numel() == B * K, "err must be [B, K]");
+  TORCH_CHECK(keys.numel() == vals.numel(), "keys/vals mismatch");
+  const float* wp = nullptr;
+  if (wts.has_value()) {
+    check_f32(*wts, "wts");
+    TORCH_CHECK(wts->numel() == B, "weights size mismatch");
+    wp = wts->data_ptr<float>();
+  }
+  mv_launch_lr_softmax_fwd(w.data_ptr<float>(), keys.data_ptr<int64_t>(),
+                           vals.data_ptr<float>(), ptr.data_ptr<int>(),
+                           labels.data_ptr<float>(), wp,
+                           err.data_ptr<float>(), loss.data_ptr<float>(),
+                           B, K, cur_stream());
+}
+
+void lr_softmax_scatter(torch::Tensor w, torch::Tensor keys,
+                        torch::Tensor vals, torch::Tensor ptr,
+                        torch::Tensor err, double lr, int64_t reg_type,
+                        double reg_coef, int64_t K) {
+  check_f32(w, "w"); check_f32(vals, "vals"); check_f32(err, "err");
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64, "keys must be int64");
+  TORCH_CHECK(ptr.scalar_type() == torch::kInt32, "ptr must be int32");
+  int64_t B = ptr.numel() - 1;
+  TORCH_CHECK(err.numel() == B * K, "err must be [B, K]");
+  mv_launch_lr_softmax_scatter(w.data_ptr<float>(), keys.data_ptr<int64_t>(),
+                               vals.data_ptr<float>(), ptr.data_ptr<int>(),
+                               err.data_ptr<float>(), (float)lr,
+                               (int)reg_type, (float)reg_coef, B, K,
+                               cur_stream());
+}
+
+void lr_ftrl_forward(torch::Tensor zn, torch::Tensor keys,
+                     torch::Tensor vals, torch::Tensor ptr,
+                     torch::Tensor labels, c10::optional<torch::Tensor> wts,
+                     torch::Tensor err, torch::Tensor loss, double alpha,
+                     double beta, double l1, double l2, int64_t K) {
+  check_f32(zn, "zn"); check_f32(vals, "vals"); check_f32(labels, "labels");
+  check_f32(err, "err"); check_f32(loss, "loss");
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64, "keys must be int64");
+  TORCH_CHECK(ptr.scalar_type() == torch::kInt32, "ptr must be int32");
+  TORCH_CHECK(K >= 1 && K <= 32,
+              "fused FTRL supports 1..32 outputs (got ", K, ")");
+  int64_t B = ptr.numel() - 1;
+  TORCH_CHECK(labels.numel() == B && loss.numel() == B, "batch mismatch");
+  TORCH_CHECK(err.numel() == B * K, "err must be [B, K]");
+  const float* wp = nullptr;
+  if (wts.has_value()) {
+    check_f32(*wts, "wts");
+    wp = wts->data_ptr<float>();
+  }
+  mv_launch_lr_ftrl_fwd(zn.data_ptr<float>(), keys.data_ptr<int64_t>(),
+                        vals.data_ptr<float>(), ptr.data_ptr<int>(),
+                        labels.data_ptr<float>(), wp, err.data_ptr<float>(),
+                        loss.data_ptr<float>(), (float)(1.0 / alpha),
+                        (float)beta, (float)l1, (float)l2, B, K,
+                        cur_stream());
+}
+
+void lr_ftrl_scatter(torch::Tensor zn, torch::Tensor keys,
+                     torch::Tensor vals, torch::Tensor ptr,
+                     torch::Tensor err, double alpha, double beta, double l1,
+                     double l2, int64_t K) {
+  check_f32(zn, "zn"); check_f32(vals, "vals"); check_f32(err, "err");
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64, "keys must be int64");
+  TORCH_CHECK(ptr.scalar_type() == torch::kInt32, "ptr must be int32");
+  int64_t B = ptr.numel() - 1;
+  TORCH_CHECK(err.numel() == B * K, "err must be [B, K]");
+  mv_launch_lr_ftrl_scatter(zn.data_ptr<float>(), keys.data_ptr<int64_t>(),
+                            vals.data_ptr<float>(), ptr.data_ptr<int>(),
+                            err.data_ptr<float>(), (float)(1.0 / alpha),
+                            (float)beta, (float)l1, (float)l2, B, K,
+                            cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -394,6 +561,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K13 fused: per-sample CSR dot + sigmoid + error/loss");
   m.def("lr_sigmoid_scatter", &lr_sigmoid_scatter,
         "K14 fused: w[keys] -= lr*(vals*err + reg), atomic");
+  m.def("lr_softmax_forward", &lr_softmax_forward,
+        "K13 softmax fused: per-sample CSR K-class logits + softmax + "
+        "err/loss (objective.cpp:193-230)");
+  m.def("lr_softmax_scatter", &lr_softmax_scatter,
+        "K14 softmax fused: w[key*K+k] -= lr*(val*err_k + reg), atomic");
+  m.def("lr_ftrl_forward", &lr_ftrl_forward,
+        "FTRL fused forward: reconstruct w from (z|n), sigmoid, err/loss "
+        "(objective.cpp:250-345)");
+  m.def("lr_ftrl_scatter", &lr_ftrl_scatter,
+        "FTRL fused z/n state update (updater.cpp:79-101 applied to the "
+        "local chunk buffer)");
   m.def("row_scatter_adagrad", &row_scatter_adagrad,
         "K15: keyed adagrad update on owned shard rows "
         "(assume_unique=True uses plain stores)",

@@ -202,8 +202,9 @@ class MatrixTable(Table):
         return ids - self.row_offset
 
     def _gather_local(self, local_ids: torch.Tensor) -> torch.Tensor:
-        """K6 on the owned shard."""
-        if self.shard.is_cuda:
+        """K6 on the owned shard (HIP kernel for the f32 hot path;
+        other dtypes use torch indexing — parity, not headline)."""
+        if self.shard.is_cuda and self.dtype == torch.float32:
             from .. import ops
             return ops.module(required=True).row_gather(self.shard, local_ids)
         return self.shard[local_ids]
@@ -239,7 +240,7 @@ class MatrixTable(Table):
             opt = option or _AO()
             gsq = self.updater.g_sqr.view(self.local_rows, self.num_col)
             vals2 = vals.view(-1, self.num_col)
-            if self.shard.is_cuda:
+            if self.shard.is_cuda and self.dtype == torch.float32:
                 from .. import ops
                 ops.module(required=True).row_scatter_adagrad(
                     self.shard, gsq, local_ids, vals2.contiguous(),
@@ -260,7 +261,7 @@ class MatrixTable(Table):
         CHECK(sign is not None,
               f"row-keyed Add with updater '{self.updater_type}' is not "
               "supported yet (stateful updaters need segmented row update)")
-        if self.shard.is_cuda:
+        if self.shard.is_cuda and self.dtype == torch.float32:
             from .. import ops
             ops.module(required=True).row_scatter_add(
                 self.shard, local_ids, vals.contiguous(), sign,

@@ -90,7 +90,8 @@ class Updater:
         stream the updated shard into ``out`` in the same pass — saves
         the Get's shard re-read. Every updater has a fused HIP kernel;
         the CPU fallback composes update + copy."""
-        if self.shard.is_cuda:
+        if self.shard.is_cuda and self.shard.dtype in (torch.float32,
+                                                        torch.float64):
             _hip_ops().sgd_copy_update(self.shard, delta, out, 1.0)
         else:
             self.update(delta, option)
@@ -111,7 +112,8 @@ class SGDUpdater(Updater):
             self.shard.sub_(delta)
 
     def update_and_copy(self, delta, option, out) -> None:
-        if self.shard.is_cuda:
+        if self.shard.is_cuda and self.shard.dtype in (torch.float32,
+                                                       torch.float64):
             _hip_ops().sgd_copy_update(self.shard, delta, out, -1.0)
         else:
             self.update(delta, option)
@@ -135,7 +137,7 @@ class MomentumUpdater(Updater):
             self.shard.sub_(self.smooth_gradient)
 
     def update_and_copy(self, delta, option, out) -> None:
-        if self.shard.is_cuda:
+        if self.shard.is_cuda and self.shard.dtype == torch.float32:
             mu = option.momentum if option else 0.0
             _hip_ops().momentum_copy_update(self.shard, self.smooth_gradient,
                                             delta, out, float(mu))
@@ -164,7 +166,7 @@ class AdaGradUpdater(Updater):
             self.shard.sub_(rho * g / torch.sqrt(self.g_sqr + self.EPS))
 
     def update_and_copy(self, delta, option, out) -> None:
-        if self.shard.is_cuda:
+        if self.shard.is_cuda and self.shard.dtype == torch.float32:
             opt = option or AddOption()
             _hip_ops().adagrad_copy_update(self.shard, self.g_sqr, delta,
                                            out, float(opt.learning_rate),
@@ -204,7 +206,7 @@ class DCASGDUpdater(Updater):
         opt = option or AddOption()
         lr, lam = opt.learning_rate, opt.lambda_
         bak = self._backup(opt.worker_id)
-        if self.shard.is_cuda:
+        if self.shard.is_cuda and self.shard.dtype == torch.float32:
             _hip_ops().dcasgd_update(self.shard, bak, delta,
                                      float(lr), float(lam))
         else:
@@ -213,7 +215,7 @@ class DCASGDUpdater(Updater):
             bak.copy_(self.shard)
 
     def update_and_copy(self, delta, option, out) -> None:
-        if self.shard.is_cuda:
+        if self.shard.is_cuda and self.shard.dtype == torch.float32:
             opt = option or AddOption()
             bak = self._backup(opt.worker_id)
             _hip_ops().dcasgd_copy_update(self.shard, bak, delta, out,
@@ -240,7 +242,7 @@ class DCASGDAUpdater(DCASGDUpdater):
         opt = option or AddOption()
         lr, lam, rho = opt.learning_rate, opt.lambda_, opt.rho
         bak = self._backup(opt.worker_id)
-        if self.shard.is_cuda:
+        if self.shard.is_cuda and self.shard.dtype == torch.float32:
             _hip_ops().dcasgda_update(self.shard, bak, self.mean_sqr, delta,
                                       float(lr), float(lam), float(rho),
                                       self.EPS)
@@ -252,7 +254,7 @@ class DCASGDAUpdater(DCASGDUpdater):
             bak.copy_(self.shard)
 
     def update_and_copy(self, delta, option, out) -> None:
-        if self.shard.is_cuda:
+        if self.shard.is_cuda and self.shard.dtype == torch.float32:
             opt = option or AddOption()
             bak = self._backup(opt.worker_id)
             _hip_ops().dcasgda_copy_update(self.shard, bak, self.mean_sqr,
