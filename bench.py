@@ -27,13 +27,20 @@ import torch.distributed as dist
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=50)
-    p.add_argument("--warmup", type=int, default=10)
+    # default timed region ~1s on GPU (0.34 ms/step x 3000) so driver
+    # gpu_busy sampling can corroborate activity
+    p.add_argument("--steps", type=int, default=3000)
+    p.add_argument("--warmup", type=int, default=100)
     p.add_argument("--rows", type=int, default=1_000_000)
     p.add_argument("--cols", type=int, default=128)
     p.add_argument("--updater", type=str, default="sgd")
+    p.add_argument("--phase-steps", dest="phase_steps", type=int, default=20,
+                   help="diagnostic per-phase (add/get/barrier) timing "
+                        "steps run AFTER the contract region; 0 disables")
+    p.add_argument("--gb", type=float, default=4.0,
+                   help="sweep mode: aggregate buffer size in GiB")
     p.add_argument("--app", type=str, default="matrix",
-                   choices=["matrix", "wordembedding", "logreg"])
+                   choices=["matrix", "wordembedding", "logreg", "sweep"])
     p.add_argument("--vocab", type=int, default=1_000_000)
     p.add_argument("--dim", type=int, default=200)
     p.add_argument("--block-words", dest="block_words", type=int,
@@ -58,6 +65,9 @@ def main():
     if args.app == "logreg":
         from multiverso_amd.apps.logreg.bench import run_bench
         run_bench(args)
+        return
+    if args.app == "sweep":
+        run_sweep(args)
         return
 
     mv.init(sync=True)
@@ -104,6 +114,35 @@ def main():
     ms_per_step = elapsed / args.steps * 1e3
     updates_per_sec = n * args.rows * args.cols / (elapsed / args.steps)
 
+    # diagnostic per-phase breakdown AFTER the contract region (VERDICT
+    # r1 #2: make the first 8-GPU run diagnosable — at N>1 add =
+    # reduce-scatter + updater kernel, get = all-gather; a slow phase
+    # points straight at the collective vs the kernel)
+    phases = None
+    if args.phase_steps > 0:
+        def timed(fn):
+            mv.barrier()
+            if cuda:
+                torch.cuda.synchronize()
+            t = time.perf_counter()
+            for _ in range(args.phase_steps):
+                fn()
+            if cuda:
+                torch.cuda.synchronize()
+            dt = time.perf_counter() - t
+            tt = torch.tensor([dt], dtype=torch.float64)
+            if dist.is_initialized() and n > 1:
+                if mv.Zoo.get().backend == "nccl":
+                    tt = tt.to(device)
+                dist.all_reduce(tt, op=dist.ReduceOp.MAX)
+            return float(tt[0]) / args.phase_steps * 1e3
+
+        phases = {
+            "add_ms": timed(lambda: (table.add(delta), table.flush())),
+            "get_ms": timed(lambda: table.get(out=out)),
+            "barrier_ms": timed(mv.barrier),
+        }
+
     if rank == 0:
         print(json.dumps({
             "metric": "MatrixTable Add+Get updates/sec (whole node)",
@@ -126,6 +165,70 @@ def main():
                 "parallelism": f"ps-sharded dp{n} (reduce-scatter/all-gather "
                                "over xGMI)",
             },
+            "phases": phases,
+        }), flush=True)
+    mv.shutdown()
+
+
+def run_sweep(args):
+    """BASELINE config 5: the 4 GB sync-SGD RCCL bucket sweep over xGMI
+    (the reference AllreduceEngine's 4096-byte switch,
+    allreduce_engine.cpp:35, as an empirically swept bucket size).
+    Times MV_Aggregate at a ladder of bucket_mb values; rank 0 prints
+    the full curve plus the contract JSON line for the best bucket."""
+    import multiverso_amd as mv
+    mv.init(sync=True)
+    n = mv.size()
+    rank = mv.rank()
+    device = mv.Zoo.get().device
+    cuda = device.type == "cuda"
+    gb = args.gb if cuda else min(args.gb, 0.0625)
+    numel = int(gb * (1 << 30) / 4)
+    buf = torch.empty(numel, dtype=torch.float32, device=device)
+    buf.uniform_(-1, 1)
+    iters = max(args.steps // 1000, 3)
+    curve = []
+    for bucket_mb in [0, 1, 2, 4, 8, 16, 32, 64, 128, 256, 512]:
+        for _ in range(2):   # warmup
+            mv.aggregate(buf, bucket_mb=bucket_mb)
+        mv.barrier()
+        if cuda:
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            mv.aggregate(buf, bucket_mb=bucket_mb)
+        if cuda:
+            torch.cuda.synchronize()
+        mv.barrier()
+        dt = (time.perf_counter() - t0) / iters
+        t = torch.tensor([dt], dtype=torch.float64)
+        if dist.is_initialized() and n > 1:
+            if mv.Zoo.get().backend == "nccl":
+                t = t.to(device)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        dt = float(t[0])
+        # all-reduce moves 2*(n-1)/n * bytes per GPU; report algo bw
+        bus_gb = (numel * 4 / 1e9) * (2 * (n - 1) / n) / dt if n > 1 else 0
+        curve.append({"bucket_mb": bucket_mb, "ms": dt * 1e3,
+                      "bus_GBps_per_gpu": bus_gb})
+        if rank == 0:
+            print(f"# bucket_mb={bucket_mb:4d}  {dt * 1e3:8.2f} ms  "
+                  f"{bus_gb:7.1f} GB/s", flush=True)
+    best = min(curve, key=lambda c: c["ms"])
+    if rank == 0:
+        print(json.dumps({
+            "metric": "MV_Aggregate 4GB all-reduce bus bandwidth",
+            "value": best["bus_GBps_per_gpu"],
+            "unit": "GB/s/gpu",
+            "n_gpus": n, "steps": iters, "warmup": 2,
+            "ms_per_step": best["ms"],
+            "higher_is_better": True, "scaling": "strong",
+            "vs_baseline": None, "dtype": "fp32", "data": "synthetic",
+            "config": {"model": f"aggregate {gb} GiB fp32",
+                       "global_batch": None, "seq_len": None,
+                       "parallelism": f"allreduce dp{n} over xGMI",
+                       "best_bucket_mb": best["bucket_mb"]},
+            "sweep": curve,
         }), flush=True)
     mv.shutdown()
 
