@@ -344,8 +344,19 @@ class AsyncEngine:
     # server side (the reference's server actor thread, server.cpp:36-58)
     # ------------------------------------------------------------------
     def _table(self, tid: int):
+        """Resolve a table id, waiting out the creation race: requests
+        are served on arrival, and a fast worker can legally send before
+        this rank's constructor finished (the reference's table
+        registration round-trip hid the same window)."""
+        import time as _time
+        deadline = _time.monotonic() + 120.0
         t = self.zoo._tables.get(tid)
+        while t is None and _time.monotonic() < deadline:
+            _time.sleep(0.001)
+            t = self.zoo._tables.get(tid)
         CHECK(t is not None, f"async request for unknown table {tid}")
+        CHECK(t._ready.wait(timeout=120.0),
+              f"table {tid} never finished construction")
         return t
 
     def _serve_loop(self) -> None:

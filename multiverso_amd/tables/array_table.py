@@ -45,6 +45,7 @@ class ArrayTable(Table):
         # single-rank Add deferral + Add/Get fusion — same mechanism and
         # semantics as MatrixTable (see matrix_table.py)
         self._deferred = None  # (delta, option, delta._version)
+        self._ready.set()
 
     def _check_deferred(self, d) -> None:
         CHECK(d[0]._version == d[2],

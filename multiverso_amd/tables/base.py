@@ -48,6 +48,11 @@ class Table:
             updater_type = get_flag("updater_type")
         self.updater_type = updater_type
         self.updater = None  # created by subclass once the shard exists
+        # async mode: a fast worker's request can arrive before THIS
+        # rank finished constructing the table (the reference's
+        # RegisterTable round-trip subsumed this); the server thread
+        # waits on _ready before touching the shard
+        self._ready = threading.Event()
         self._pending: List[Handle] = []
         # serializes the async server thread against local worker ops on
         # the same shard (the reference's one-consumer-thread-per-actor

@@ -28,6 +28,7 @@ class KVTable(Table):
         super().__init__(updater_type="default")
         self._store: Dict[int, float] = {}   # server-side shard (my keys)
         self._mirror: Dict[int, float] = {}  # worker-side local mirror
+        self._ready.set()
 
     def _owner(self, key: int) -> int:
         return key % self.zoo.num_servers

@@ -68,6 +68,7 @@ class MatrixTable(Table):
         # headline config). Any other table op materializes the Add
         # first (flush).
         self._deferred = None  # (delta, option, delta._version)
+        self._ready.set()
 
     def _check_deferred(self, d) -> None:
         CHECK(d[0]._version == d[2],

@@ -199,3 +199,67 @@ def test_array_uneven_shards_ws3():
 
 def test_matrix_whole_and_rows_ws4():
     run_dist(_matrix_whole_and_rows, 4)
+
+
+# ---- world_size=8: the exact rank count of the driver's 8-GPU node.
+# One combined collective-path test (array oracle + keyed rows + sparse
+# stale get + aggregate) keeps the 8-process spawn cost down.
+
+def _all_paths_ws8(rank, world):
+    import multiverso_amd as mv
+    mv.init(sync=True)
+    # array BSP oracle (test_array_table.cpp:31-42 pattern)
+    a = mv.ArrayTable(19)            # uneven: 2x7 + 5 at ws8
+    for it in range(2):
+        a.add(torch.full((19,), 2.0)).wait()
+        got = a.get()
+        assert torch.equal(got, torch.full((19,), 2.0 * (it + 1) * world))
+    # matrix whole + keyed (every rank different rows; some collide)
+    m = mv.MatrixTable(21, 3)        # 2 rows x 7 + 7 on the last
+    m.add(torch.ones(21, 3))
+    m.add_rows([rank, 20 - rank, 10], torch.ones(3, 3))
+    got = m.get_rows(list(range(21)))
+    expect = torch.full((21, 3), float(world))
+    for r in range(world):
+        expect[r] += 1.0
+        expect[20 - r] += 1.0
+    expect[10] += world
+    assert torch.equal(got, expect), (rank, got[:, 0])
+    # sparse stale-filtered get
+    s = mv.SparseMatrixTable(17, 2)
+    cache = torch.zeros(17, 2)
+    assert s.get_into(cache) == 17
+    assert s.get_into(cache) == 0
+    if rank == 3:
+        s.add_rows([0, 16], torch.ones(2, 2))
+    else:
+        s.add_rows([], torch.zeros(0, 2))
+    assert s.get_into(cache) == 2
+    assert torch.equal(cache[16], torch.ones(2))
+    # aggregate
+    x = torch.ones(3)
+    mv.aggregate(x)
+    assert torch.equal(x, torch.full((3,), float(world)))
+    mv.shutdown()
+
+
+def test_all_collective_paths_ws8():
+    run_dist(_all_paths_ws8, 8)
+
+
+def _async_ws8(rank, world):
+    import multiverso_amd as mv
+    mv.init()
+    t = mv.ArrayTable(16)
+    for _ in range(rank + 1):        # every rank a DIFFERENT op count
+        t.add(torch.ones(16))
+        t.get()
+    mv.barrier()
+    got = t.get()
+    total = sum(r + 1 for r in range(world))
+    assert torch.equal(got, torch.full((16,), float(total))), (rank, got)
+    mv.shutdown()
+
+
+def test_async_ws8():
+    run_dist(_async_ws8, 8)

@@ -254,3 +254,40 @@ def test_huffman_file_roundtrip(tmp_path):
         assert x.code == y.code and x.point == y.point
     first = open(p).read().splitlines()
     assert first[0] == "5" and first[1].startswith("a ")
+
+
+def _w2v_hogwild_ws8(rank, world):
+    """8-rank convergence oracle for the pull-train-push staleness x
+    hogwild interaction: 8 workers train the same
+    even->odd structure concurrently in TRUE ASYNC mode (keyed get/add
+    served on arrival); after the drain barrier the embeddings must
+    still separate positives from mismatched pairs."""
+    import torch
+    import multiverso_amd as mv
+    from multiverso_amd.apps.wordembedding.model import (WordEmbedding,
+                                                         WordEmbeddingOption)
+    mv.init()   # async mode — the real ASGD w2v
+    torch.manual_seed(rank)
+    opt = WordEmbeddingOption(embedding_size=16, window=1, negative_num=3,
+                              init_learning_rate=0.08,
+                              total_words=50_000_000, seed=3 + rank)
+    model = WordEmbedding(opt, [100] * 12)
+    words = torch.stack([torch.arange(0, 12, 2).repeat(20),
+                         torch.arange(1, 12, 2).repeat(20)], dim=1).view(-1)
+    sids = torch.arange(words.numel()) // 10
+    for _ in range(4):
+        model.train_block(words, sids)
+        model.sync_word_count()
+    mv.barrier()
+    inp = model.input_table.get()
+    out = model.output_table.get()
+    evens = torch.arange(0, 12, 2)
+    pos = torch.sigmoid((inp[evens] * out[evens + 1]).sum(1)).mean()
+    wrong = torch.sigmoid((inp[evens] * out[evens.roll(1) + 1]).sum(1)).mean()
+    assert float(pos) - float(wrong) > 0.15, (rank, float(pos), float(wrong))
+    mv.shutdown()
+
+
+def test_w2v_hogwild_convergence_ws8():
+    from conftest import run_dist
+    run_dist(_w2v_hogwild_ws8, 8, timeout=240)
