@@ -73,6 +73,19 @@ class Logger:
     def fatal(self, msg: str) -> None:
         self._emit(LogLevel.FATAL, msg)
         if self.kill_fatal:
+            # Fail-fast parity with the reference (CHECK aborts the MPI
+            # job, log.h:10-13): tear down the process group so peer
+            # ranks blocked in a collective fail promptly instead of
+            # hanging out their 300 s timeout. destroy is best-effort —
+            # a rank inside a collective cannot always be interrupted,
+            # but dead TCP connections (gloo) / closed communicators
+            # surface the failure to peers.
+            try:
+                import torch.distributed as dist
+                if dist.is_initialized():
+                    dist.destroy_process_group()
+            except Exception:
+                pass
             raise FatalError(msg)
 
 

@@ -127,6 +127,63 @@ class SparseMatrixTable(MatrixTable):
         return (torch.cat(outs) if outs else
                 torch.empty(0, dtype=self.dtype, device=device))
 
+    # ---- pipelined (double-buffered) stale get ----
+    def prefetch_into(self, cache: torch.Tensor) -> "Handle":
+        """The unified Matrix's ``is_pipeline`` capability
+        (matrix.cpp:384-420: doubled up_to_date bitmap so the next
+        block's Get proceeds while the current block computes): issue
+        the stale-row exchange NOW, overlap with compute, apply to
+        ``cache`` on ``wait()``. Rows are marked fresh at issue time —
+        adds landing between issue and wait re-invalidate them, so the
+        next get re-pulls (the doubled-bitmap semantics). Collective:
+        every rank must call. The value exchange is dense (the
+        SparseFilter's per-segment packing is synchronous by nature);
+        ``handle.rows`` carries the incoming row count after wait()."""
+        from ..comm import Handle, exchange_sizes
+        CHECK(cache.shape == (self.num_row, self.num_col),
+              "cache must be the full table shape")
+        self.flush()
+        if self.zoo.size == 1:
+            n = self.get_into(cache)
+            h = Handle()
+            h.rows = n
+            return h
+        nw = self.zoo.num_workers
+        with monitor("worker.sparse_prefetch"):
+            stale_lists = [(~self.up_to_date[w]).nonzero().reshape(-1)
+                           for w in range(nw)]
+            send_rows = [int(s.numel()) for s in stale_lists]
+            recv_rows = exchange_sizes(send_rows)
+            all_ids = (torch.cat(stale_lists) + self.row_offset
+                       if sum(send_rows) else
+                       torch.empty(0, dtype=torch.int64, device=self.device))
+            got_ids = torch.empty(sum(recv_rows), dtype=torch.int64,
+                                  device=self.device)
+            w1 = dist.all_to_all_single(got_ids, all_ids, recv_rows,
+                                        send_rows, async_op=True)
+            served = (self._gather_local(torch.cat(stale_lists))
+                      if sum(send_rows) else
+                      torch.empty(0, self.num_col, dtype=self.dtype,
+                                  device=self.device))
+            got_vals = torch.empty(sum(recv_rows) * self.num_col,
+                                   dtype=self.dtype, device=self.device)
+            w2 = dist.all_to_all_single(
+                got_vals, served.reshape(-1),
+                [r * self.num_col for r in recv_rows],
+                [s * self.num_col for s in send_rows], async_op=True)
+            for w in range(nw):
+                if send_rows[w]:
+                    self.up_to_date[w, stale_lists[w]] = True
+
+        def finish() -> None:
+            w1.wait()
+            w2.wait()
+            if got_ids.numel():
+                cache[got_ids] = got_vals.view(-1, self.num_col)
+        h = Handle(None, finish)
+        h.rows = sum(recv_rows)
+        return self._track(h)
+
     # ---- stale-filtered whole-table get ----
     def get_into(self, cache: torch.Tensor) -> int:
         """Overwrite only the rows of ``cache`` that are stale for this

@@ -263,3 +263,25 @@ def _async_ws8(rank, world):
 
 def test_async_ws8():
     run_dist(_async_ws8, 8)
+
+
+def _kv_large(rank, world):
+    """KV with a large key set (tensor exchange, not pickle): 20k keys
+    per rank, half overlapping."""
+    import multiverso_amd as mv
+    mv.init(sync=True)
+    t = mv.KVTable()
+    base = 10_000 * rank
+    keys = list(range(base, base + 20_000))
+    t.add(keys, [1.0] * len(keys))
+    got = t.get([5_000, 10_000 + 5_000 * (world - 1), 999_999])
+    # overlap region [10000, 20000) got world contributions at ws2
+    assert got[5_000] == 1.0
+    if world == 2:
+        assert got[15_000] == 2.0
+    assert got[999_999] == 0.0
+    mv.shutdown()
+
+
+def test_kv_large_keyset():
+    run_dist(_kv_large, 2)

@@ -231,3 +231,43 @@ def _sparse_filter_wire(rank, world):
 def test_sparse_filter_wire_dist():
     from conftest import run_dist
     run_dist(_sparse_filter_wire, 2)
+
+
+def _sparse_prefetch_dist(rank, world):
+    """Pipelined stale get (the unified Matrix's is_pipeline,
+    matrix.cpp:384-420): prefetch overlaps with an add; rows marked
+    fresh at issue are re-invalidated by the interleaved add, so the
+    NEXT get re-pulls them."""
+    import multiverso_amd as mv
+    import torch
+    mv.init(sync=True)
+    t = mv.SparseMatrixTable(9, 3)
+    cache = torch.zeros(9, 3)
+    h = t.prefetch_into(cache)      # issue the initial full pull
+    # interleaved keyed add while the exchange is in flight
+    if rank == 0:
+        t.add_rows([1], torch.ones(1, 3))
+    else:
+        t.add_rows([], torch.zeros(0, 3))
+    h.wait()
+    assert h.rows == 9
+    # prefetch snapshotted PRE-add values; the add re-invalidated row 1
+    n = t.get_into(cache)
+    assert n == 1, n
+    assert torch.equal(cache[1], torch.ones(3))
+    assert t.get_into(cache) == 0
+    mv.shutdown()
+
+
+def test_sparse_prefetch_dist():
+    run_dist(_sparse_prefetch_dist, 2)
+
+
+def test_sparse_prefetch_local(env):
+    import torch
+    t = mv.SparseMatrixTable(6, 2)
+    cache = torch.zeros(6, 2)
+    h = t.prefetch_into(cache)
+    h.wait()
+    assert h.rows == 6
+    assert t.get_into(cache) == 0
