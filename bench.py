@@ -47,6 +47,11 @@ def parse_args():
                    default=500_000)
     p.add_argument("--use-adagrad", dest="use_adagrad", action="store_true",
                    help="wordembedding: per-element AdaGrad mode")
+    p.add_argument("--objective", type=str, default="sigmoid",
+                   choices=["sigmoid", "softmax", "ftrl"],
+                   help="logreg: objective (fused kernel per objective)")
+    p.add_argument("--classes", type=int, default=10,
+                   help="logreg softmax: number of classes")
     return p.parse_args()
 
 

@@ -30,15 +30,20 @@ def run_bench(args):
     if not cuda:
         input_size, minibatch = min(input_size, 100_000), min(minibatch, 256)
 
+    objective = getattr(args, "objective", "sigmoid")
+    classes = getattr(args, "classes", 10) if objective == "softmax" else 1
     cfg = LogRegConfig(input_size=input_size, minibatch_size=minibatch,
-                       use_ps=True, updater_type="adagrad",
+                       use_ps=True,
+                       updater_type="sgd" if objective == "ftrl"
+                       else "adagrad",
+                       objective_type=objective, output_size=classes,
                        sync_frequency=sync_freq, learning_rate=0.05,
                        show_time_per_sample=0)
     model = PSModel(cfg, device)
 
     n_chunks = args.warmup + args.steps
     batches, _ = synthetic_batches(input_size, n_chunks * sync_freq,
-                                   minibatch, nnz=nnz,
+                                   minibatch, nnz=nnz, output_size=classes,
                                    seed=31 + rank, device=device)
     chunks = [batches[i * sync_freq:(i + 1) * sync_freq]
               for i in range(n_chunks)]
@@ -82,7 +87,8 @@ def run_bench(args):
             "data": "synthetic",
             "config": {
                 "model": f"sparse logistic regression {input_size} features "
-                         f"nnz={nnz} adagrad",
+                         f"nnz={nnz} {objective}"
+                         + (f" K={classes}" if classes > 1 else ""),
                 "global_batch": n * minibatch * sync_freq,
                 "seq_len": None,
                 "parallelism": f"ps-sharded dp{n} (row all-to-all over xGMI)",

