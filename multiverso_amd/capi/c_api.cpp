@@ -32,6 +32,13 @@ struct Handle {
 
 py::module_ mv() { return py::module_::import("multiverso_amd"); }
 
+// Some entry points are legal BEFORE MV_Init (SetFlag, NetBind/Connect
+// — reference multiverso.cpp:48-68); they must boot the embedded
+// interpreter themselves.
+void ensure_interp() {
+  if (!Py_IsInitialized()) g_interp = new py::scoped_interpreter();
+}
+
 py::array_t<float> wrap(float* data, int size) {
   // zero-copy view of caller memory
   return py::array_t<float>({(py::ssize_t)size}, {sizeof(float)}, data,
@@ -48,9 +55,7 @@ py::object to_tensor(float* data, int size) {
 extern "C" {
 
 void MV_Init(int* argc, char* argv[]) {
-  if (!Py_IsInitialized()) {
-    g_interp = new py::scoped_interpreter();
-  }
+  ensure_interp();
   py::gil_scoped_acquire gil;
   py::list args;
   if (argc && argv)
@@ -181,6 +186,53 @@ void MV_AddAsyncMatrixTableByRows(TableHandler handler, float* data, int size,
   // keyed adds complete within the call in the collective design; the
   // async variant is equivalent (reference fire-and-forget semantics).
   MV_AddMatrixTableByRows(handler, data, size, row_ids, row_ids_n);
+}
+
+// ---- extensions beyond the reference C API (see c_api.h) ----
+
+int MV_Rank() {
+  py::gil_scoped_acquire gil;
+  return mv().attr("rank")().cast<int>();
+}
+
+int MV_Size() {
+  py::gil_scoped_acquire gil;
+  return mv().attr("size")().cast<int>();
+}
+
+int MV_NumServers() {
+  py::gil_scoped_acquire gil;
+  return mv().attr("servers_num")().cast<int>();
+}
+
+void MV_Aggregate(float* data, int size) {
+  py::gil_scoped_acquire gil;
+  // in-place: aggregate() sums into the tensor view of caller memory
+  mv().attr("aggregate")(to_tensor(data, size));
+}
+
+void MV_SetFlag(const char* key, const char* value) {
+  ensure_interp();
+  py::gil_scoped_acquire gil;
+  mv().attr("set_flag")(std::string(key), std::string(value));
+}
+
+int MV_NetBind(int rank, const char* endpoint) {
+  ensure_interp();
+  py::gil_scoped_acquire gil;
+  return mv().attr("net_bind")(rank, std::string(endpoint)).cast<bool>()
+             ? 1 : 0;
+}
+
+int MV_NetConnect(int* ranks, char* endpoints[], int n) {
+  ensure_interp();
+  py::gil_scoped_acquire gil;
+  py::list rs, eps;
+  for (int i = 0; i < n; ++i) {
+    rs.append(ranks[i]);
+    eps.append(std::string(endpoints[i]));
+  }
+  return mv().attr("net_connect")(rs, eps).cast<bool>() ? 1 : 0;
 }
 
 }  // extern "C"

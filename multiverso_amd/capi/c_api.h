@@ -66,6 +66,26 @@ DllExport void MV_AddAsyncMatrixTableByRows(TableHandler handler, float* data,
                                             int size, int row_ids[],
                                             int row_ids_n);
 
+/* ---- Extensions beyond the reference c_api.h (the reference exposed
+ * these only through its C++ API, include/multiverso/multiverso.h:9-68;
+ * they are provided here so the C++ header multiverso.hpp — and any
+ * other FFI host — reaches the full MV_* surface). The exact reference
+ * symbol set above is unchanged. ---- */
+
+DllExport int MV_Rank();
+DllExport int MV_Size();
+DllExport int MV_NumServers();
+/* In-place float sum-allreduce across all ranks (MV_Aggregate,
+ * src/multiverso.cpp:53-56): rcclAllReduce over xGMI on GPU nodes. */
+DllExport void MV_Aggregate(float* data, int size);
+/* Programmatic flag set (MV_SetFlag, src/multiverso.cpp:48-51). */
+DllExport void MV_SetFlag(const char* key, const char* value);
+/* Explicit launcher-free rendezvous (MV_NetBind/MV_NetConnect,
+ * src/multiverso.cpp:58-68): declare this process's rank, then provide
+ * the full rank->endpoint map; call both BEFORE MV_Init. */
+DllExport int MV_NetBind(int rank, const char* endpoint);
+DllExport int MV_NetConnect(int* ranks, char* endpoints[], int n);
+
 #ifdef __cplusplus
 }
 #endif

@@ -35,8 +35,9 @@ def free_port() -> int:
     return port
 
 
-def _dist_entry(rank: int, fn, world_size: int, port: int, backend: str):
-    if backend == "gloo":
+def _dist_entry(rank: int, fn, world_size: int, port: int, backend: str,
+                gpu_share: bool = False):
+    if backend == "gloo" and not gpu_share:
         # these are CPU-tier tests: mask any GPU so world_size ranks on a
         # 1-GPU host don't map LOCAL_RANK -> missing cuda devices
         os.environ["HIP_VISIBLE_DEVICES"] = ""
@@ -45,12 +46,17 @@ def _dist_entry(rank: int, fn, world_size: int, port: int, backend: str):
     os.environ["MASTER_PORT"] = str(port)
     os.environ["RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world_size)
-    os.environ["LOCAL_RANK"] = str(rank)
+    # gpu_share: every rank maps the SAME single GPU (cuda:0) — used to
+    # exercise the host-staged async engine with CUDA shards on 1-GPU
+    # boxes (the collective/RCCL plane needs real multi-GPU and is the
+    # driver's job)
+    os.environ["LOCAL_RANK"] = "0" if gpu_share else str(rank)
     os.environ["MV_BACKEND"] = backend
     fn(rank, world_size)
 
 
-def run_dist(fn, world_size: int = 2, backend: str = "gloo", timeout: int = 120):
+def run_dist(fn, world_size: int = 2, backend: str = "gloo",
+             timeout: int = 120, gpu_share: bool = False):
     """Run ``fn(rank, world_size)`` in world_size fresh processes over gloo.
 
     This is the rebuild's analog of the reference's `mpirun -np N` CLI
@@ -58,5 +64,6 @@ def run_dist(fn, world_size: int = 2, backend: str = "gloo", timeout: int = 120)
     """
     import torch.multiprocessing as mp
     port = free_port()
-    mp.start_processes(_dist_entry, args=(fn, world_size, port, backend),
+    mp.start_processes(_dist_entry,
+                       args=(fn, world_size, port, backend, gpu_share),
                        nprocs=world_size, join=True, start_method="spawn")

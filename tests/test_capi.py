@@ -112,3 +112,63 @@ def test_c_host_roundtrip_sanitized(tmp_path_factory):
                        timeout=600)
     assert r.returncode == 0, (r.returncode, r.stdout[-500:], r.stderr[-2000:])
     assert "CAPI_OK" in r.stdout
+
+
+CPP_HOST = r"""
+#include <cstdio>
+#include <vector>
+#include "multiverso.hpp"
+
+int main() {
+  multiverso::SetCMDFlag("sync", "true");
+  multiverso::Init();
+  if (multiverso::Rank() != 0 || multiverso::Size() != 1) return 2;
+  if (multiverso::NumWorkers() != 1 || multiverso::NumServers() != 1)
+    return 2;
+  multiverso::Barrier();
+
+  multiverso::ArrayTableHandler arr(8);
+  std::vector<float> ones(8, 1.0f), out;
+  arr.Add(ones.data());
+  arr.Get(out);
+  for (float v : out) if (v != 1.0f) return 3;
+
+  std::vector<float> agg(4, 2.0f);
+  multiverso::Aggregate(agg);          // world 1: identity
+  if (agg[0] != 2.0f) return 4;
+
+  multiverso::MatrixTableHandler mat(6, 2);
+  std::vector<float> rv = {5, 5, 7, 7}, ro(4);
+  std::vector<int> rows = {1, 4};
+  mat.AddByRows(rv.data(), rows);
+  mat.GetByRows(ro.data(), rows);
+  if (ro[0] != 5.0f || ro[2] != 7.0f) return 5;
+  std::vector<float> whole(12);
+  mat.GetAll(whole.data());
+  if (whole[2] != 5.0f || whole[8] != 7.0f) return 6;
+
+  multiverso::ShutDown();
+  std::printf("CPPAPI_OK\n");
+  return 0;
+}
+"""
+
+
+def test_cpp_host_roundtrip(capi_so, tmp_path_factory):
+    """The reference's public C++ API surface (multiverso.h:9-68) as a
+    native C++ host program over multiverso.hpp."""
+    tmp = tmp_path_factory.mktemp("cppapi")
+    src = tmp / "host.cpp"
+    src.write_text(CPP_HOST)
+    exe = tmp / "host"
+    inc = os.path.join(REPO, "multiverso_amd", "capi")
+    subprocess.run(
+        ["g++", "-O0", "-std=c++17", str(src), f"-I{inc}", f"-L{inc}",
+         "-lmultiverso_amd", "-o", str(exe)], check=True)
+    env = dict(os.environ)
+    env["LD_LIBRARY_PATH"] = inc + ":" + env.get("LD_LIBRARY_PATH", "")
+    env["PYTHONPATH"] = REPO + ":" + env.get("PYTHONPATH", "")
+    r = subprocess.run([str(exe)], env=env, capture_output=True, text=True,
+                       timeout=300)
+    assert r.returncode == 0, (r.returncode, r.stdout, r.stderr)
+    assert "CPPAPI_OK" in r.stdout
