@@ -131,7 +131,8 @@ class KVTable(Table):
                                   for kk in mine.tolist()],
                                  dtype=torch.float64)
             have = torch.tensor([kk in self._store
-                                 for kk in mine.tolist()])
+                                 for kk in mine.tolist()],
+                                dtype=torch.bool)
             ak, av = self._allgather_kv(mine[have], mvals[have])
             merged = dict(zip(ak.tolist(), av.tolist()))
         else:
