@@ -37,6 +37,9 @@ class MVTorchParamManager:
         if not is_master_worker():
             init = torch.zeros_like(init)
         self.table.add(init)
+        # async mode: make the master's init visible to all workers
+        # before the first read (no-op cost under BSP sync)
+        Zoo.get().barrier()
         self._last = self.table.get().clone()
         self._unflatten(self._last)
 

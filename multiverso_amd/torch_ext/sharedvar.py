@@ -26,6 +26,10 @@ class MVSharedTensor:
         if not is_master_worker():
             init = torch.zeros_like(init)
         self._table.add(init)
+        # async mode: the master's init must be visible to every worker
+        # before anyone reads (no-op cost in sync mode — see handlers)
+        from ..zoo import Zoo
+        Zoo.get().barrier()
         self._last = self._table.get().clone()
         with torch.no_grad():
             self.tensor.reshape(-1).copy_(self._last.to(tensor.device))
