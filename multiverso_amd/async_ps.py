@@ -43,6 +43,7 @@ from typing import Dict, List, Optional
 import torch
 import torch.distributed as dist
 
+from .dashboard import monitor
 from .log import CHECK, log
 from .updaters import AddOption
 
@@ -405,21 +406,25 @@ class AsyncEngine:
             option = _opt_from(ot)
 
         if op == OP_ADD:
-            table._server_apply_chunk(vals.to(table.device), option)
+            with monitor("server.process_add"):
+                table._server_apply_chunk(vals.to(table.device), option)
             if want_ack:
                 dist.send(torch.zeros(1, dtype=torch.int64), src,
                           group=self.rep)
         elif op == OP_GET:
-            out = torch.empty(n_vals, dtype=table.dtype)
-            table._server_read_chunk_into(out)
+            with monitor("server.process_get"):
+                out = torch.empty(n_vals, dtype=table.dtype)
+                table._server_read_chunk_into(out)
             dist.send(out, src, group=self.rep)
         elif op == OP_ADD_ROWS:
-            local = (keys - table.row_offset).to(table.device)
-            table._server_add_rows(
-                local, vals.view(n_keys, -1).to(table.device), option)
+            with monitor("server.process_add"):
+                local = (keys - table.row_offset).to(table.device)
+                table._server_add_rows(
+                    local, vals.view(n_keys, -1).to(table.device), option)
         elif op == OP_GET_ROWS:
-            local = (keys - table.row_offset).to(table.device)
-            got = table._server_get_rows(local)
+            with monitor("server.process_get"):
+                local = (keys - table.row_offset).to(table.device)
+                got = table._server_get_rows(local)
             dist.send(got.reshape(-1).cpu().contiguous(), src,
                       group=self.rep)
         elif op == OP_KV_ADD:
