@@ -264,3 +264,63 @@ def _param_manager_async(rank, world):
 
 def test_param_manager_async():
     run_dist(_param_manager_async, 2)
+
+
+def _soak(rank, world):
+    """Protocol soak: every rank issues a random interleaving of
+    whole-table adds/gets, keyed adds/gets, KV adds and barriers at its
+    own pace; the final drained state must equal the exact sum of
+    everything everyone sent (counted locally, cross-checked via
+    aggregate)."""
+    import random
+    import multiverso_amd as mv
+    mv.init()
+    rng = random.Random(1234 + rank)
+    arr = mv.ArrayTable(33)            # uneven shards at ws3: 11+11+11
+    mat = mv.MatrixTable(17, 3)
+    kv = mv.KVTable()
+    my_arr = 0.0
+    my_rows = torch.zeros(17)
+    my_kv = 0.0
+    for i in range(60):
+        op = rng.randrange(6)
+        if op == 0:
+            arr.add(torch.ones(33), async_op=bool(rng.getrandbits(1)))
+            my_arr += 1.0
+        elif op == 1:
+            arr.get()
+        elif op == 2:
+            rows = rng.sample(range(17), rng.randrange(1, 5))
+            mat.add_rows(rows, torch.ones(len(rows), 3))
+            for r in rows:
+                my_rows[r] += 1.0
+        elif op == 3:
+            mat.get_rows(rng.sample(range(17), 3))
+        elif op == 4:
+            k = rng.randrange(5)
+            kv.add([k], [1.0])
+            my_kv += 1.0
+        # (no in-loop barriers: MV_Barrier is collective — same count on
+        # every rank — in the reference too; the async freedom is in the
+        # TABLE ops, which here interleave arbitrarily across ranks)
+        if rng.random() < 0.1:
+            time.sleep(0.001 * rng.random())
+    mv.barrier()
+    # exact totals: sum of per-rank contributions
+    tot = torch.tensor([my_arr])
+    mv.aggregate(tot)
+    got = arr.get()
+    assert torch.equal(got, torch.full((33,), float(tot[0]))), (rank, got)
+    rows_tot = my_rows.clone()
+    mv.aggregate(rows_tot)
+    got_rows = mat.get_rows(list(range(17)))
+    assert torch.equal(got_rows, rows_tot.unsqueeze(1).expand(17, 3)), rank
+    kv_tot = torch.tensor([my_kv])
+    mv.aggregate(kv_tot)
+    got_kv = kv.get([0, 1, 2, 3, 4])
+    assert abs(sum(got_kv.values()) - float(kv_tot[0])) < 1e-6, rank
+    mv.shutdown()
+
+
+def test_async_soak_ws3():
+    run_dist(_soak, 3, timeout=240)
