@@ -125,23 +125,29 @@ class PSModel:
             cat_keys = torch.cat([b.keys for b in batches])
             if self.cfg.input_size <= (1 << 31) - 1:
                 # keys fit int32: the dedup sort runs at half the bytes
-                union = torch.unique(cat_keys.to(torch.int32)).to(
-                    torch.int64)
+                union, inv = torch.unique(cat_keys.to(torch.int32),
+                                          return_inverse=True)
+                union = union.to(torch.int64)
             else:
-                union = torch.unique(cat_keys)
+                union, inv = torch.unique(cat_keys, return_inverse=True)
+            # the inverse IS each occurrence's local row in the pulled
+            # buffer — split per minibatch and the per-batch searchsorted
+            # (4x ~21 us/chunk on MI355X) disappears
+            sizes = [b.keys.numel() for b in batches]
+            lidxs = list(torch.split(inv, sizes))
         else:
             union = torch.empty(0, dtype=torch.int64, device=self.device)
+            lidxs = []
         pulled = self.table.get_rows(union)
         local = pulled.clone()
         total_loss = 0.0
         adagrad = self.table.updater_type == "adagrad"
         fused = self._fused_kind(local)
         if fused:
-            total_loss = self._train_chunk_fused(batches, union, local,
+            total_loss = self._train_chunk_fused(batches, lidxs, local,
                                                  fused)
         else:
-            for b in batches:
-                lidx = torch.searchsorted(union, b.keys)
+            for b, lidx in zip(batches, lidxs):
                 w_rows = local[lidx]
                 grad, loss = self.objective.gradient(b, w_rows)
                 total_loss += loss
@@ -185,7 +191,7 @@ class PSModel:
             return "ftrl"
         return ""
 
-    def _train_chunk_fused(self, batches, union, local, kind) -> float:
+    def _train_chunk_fused(self, batches, lidxs, local, kind) -> float:
         """Two HIP kernels per minibatch (fwd + scatter) instead of ~20
         torch ops; numerics match objective.gradient + scatter
         (tests/test_gpu_kernels.py). Returns summed per-batch mean
@@ -202,8 +208,7 @@ class PSModel:
         K = self.cfg.output_size
         wflat = local.view(-1)
         loss_acc = torch.zeros((), device=self.device)
-        for b in batches:
-            lidx = torch.searchsorted(union, b.keys)
+        for b, lidx in zip(batches, lidxs):
             ptr32 = b.ptr.to(torch.int32)
             B = b.size
             lossb = torch.empty(B, device=self.device)
