@@ -37,6 +37,21 @@ namespace Multiverso
         [DllImport(Lib, EntryPoint = "MV_ServerId")]
         private static extern int MV_ServerId();
 
+        [DllImport(Lib, EntryPoint = "MV_Rank")]
+        public static extern int Rank();
+
+        [DllImport(Lib, EntryPoint = "MV_Size")]
+        public static extern int Size();
+
+        [DllImport(Lib, EntryPoint = "MV_NumServers")]
+        public static extern int NumServers();
+
+        [DllImport(Lib, EntryPoint = "MV_Aggregate")]
+        public static extern void Aggregate(float[] data, int size);
+
+        [DllImport(Lib, EntryPoint = "MV_SetFlag")]
+        public static extern void SetFlag(string key, string value);
+
         [DllImport(Lib, EntryPoint = "MV_NewArrayTable")]
         private static extern void MV_NewArrayTable(int size, out IntPtr handler);
         [DllImport(Lib, EntryPoint = "MV_GetArrayTable")]

@@ -34,6 +34,11 @@ void MV_GetMatrixTableByRows(TableHandler handler, float* data, int size,
                              int row_ids[], int row_ids_n);
 void MV_AddMatrixTableByRows(TableHandler handler, float* data, int size,
                              int row_ids[], int row_ids_n);
+int MV_Rank();
+int MV_Size();
+int MV_NumServers();
+void MV_Aggregate(float* data, int size);
+void MV_SetFlag(const char* key, const char* value);
 void MV_AddAsyncMatrixTableByRows(TableHandler handler, float* data, int size,
                                   int row_ids[], int row_ids_n);
 ]]
@@ -49,6 +54,10 @@ end
 
 function mv.shutdown() lib.MV_ShutDown() end
 function mv.barrier() lib.MV_Barrier() end
+function mv.rank() return lib.MV_Rank() end
+function mv.size() return lib.MV_Size() end
+function mv.servers_num() return lib.MV_NumServers() end
+function mv.set_flag(k, v) lib.MV_SetFlag(k, tostring(v)) end
 function mv.num_workers() return lib.MV_NumWorkers() end
 function mv.worker_id() return lib.MV_WorkerId() end
 function mv.server_id() return lib.MV_ServerId() end
