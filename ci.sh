@@ -10,7 +10,7 @@ python __graft_entry__.py build
 python -c 'from multiverso_amd import capi; capi.build(verbose=True)'
 
 echo "== CPU suite (logic + single-process runtime + world_size 2/4 gloo \
-+ binding + CLI end-to-end; GPU masked so device selection and \
++ ws3/ws8 + async-PS + binding + CLI end-to-end; GPU masked so device selection and \
 multi-process semantics match the CPU tier) =="
 HIP_VISIBLE_DEVICES="" CUDA_VISIBLE_DEVICES="" \
     python -m pytest tests -q -m "not gpu"

@@ -324,3 +324,22 @@ def _soak(rank, world):
 
 def test_async_soak_ws3():
     run_dist(_soak, 3, timeout=240)
+
+
+def _reinit_cycle(rank, world):
+    """init -> async ops -> shutdown -> init -> ops again in ONE process
+    (the reference allowed re-Init; server threads and groups must tear
+    down and rebuild cleanly)."""
+    import multiverso_amd as mv
+    for cycle in range(2):
+        mv.init()
+        t = mv.ArrayTable(8)
+        t.add(torch.ones(8))
+        mv.barrier()
+        got = t.get()
+        assert torch.equal(got, torch.full((8,), float(world))), (cycle, got)
+        mv.shutdown()
+
+
+def test_async_reinit_cycle():
+    run_dist(_reinit_cycle, 2)
