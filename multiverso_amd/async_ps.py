@@ -38,7 +38,7 @@ from __future__ import annotations
 
 import datetime
 import threading
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 import torch
 import torch.distributed as dist
