@@ -32,6 +32,19 @@
 // still: 6.05/5.84 TB/s at 768 blocks vs 5.45/5.15 at 1024 (same sweep).
 #define ELEM_GRID 1024
 #define STATE_GRID 768
+// 4-stream fused Add+Get kernels prefer FAT blocks: 1024 threads x 768+
+// blocks measured 6.14 TB/s vs 5.87 at 256x768 (tools/probe_sgd2.hip
+// round-2 sweep; the 2/3-stream kernels regress at 1024 — they keep
+// BLOCK=256). Grid counts are in BLOCKS of COPY_BLOCK threads.
+#define COPY_BLOCK 1024
+#define COPY_GRID 768
+
+static inline int grid_copy(int64_t work_items) {
+  int64_t blocks = (work_items + COPY_BLOCK - 1) / COPY_BLOCK;
+  if (blocks > COPY_GRID) blocks = COPY_GRID;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
 
 typedef float v4f __attribute__((ext_vector_type(4)));
 
@@ -520,13 +533,12 @@ void mv_launch_adagrad(float* data, float* gsq, const float* delta,
 }
 
 // sign = +1 for the default (add) updater, -1 for sgd.
-// Grid 768 (3 blocks/CU): the 2-read/2-write stream measured 6.05 TB/s
-// at 768 blocks vs 5.45 at 1024+ (tools/probe_sgd.hip fused sweep) — a
-// wider grid over-subscribes the write queues.
+// Fat-block geometry (COPY_BLOCK x COPY_GRID): see the round-2 sweep
+// note at the COPY_BLOCK define.
 void mv_launch_sgd_copy(float* data, const float* delta, float* out,
                         float sign, int64_t n, hipStream_t s) {
   int64_t n4 = n / 4;
-  if (n4) k_sgd_copy_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
+  if (n4) k_sgd_copy_f4<<<grid_copy(n4), COPY_BLOCK, 0, s>>>(
       (v4f*)data, (const v4f*)delta, (v4f*)out, sign, n4);
   int64_t tail = n - n4 * 4;
   if (tail) k_sgd_copy_tail<<<1, 64, 0, s>>>(data, delta, out, sign,
@@ -536,7 +548,7 @@ void mv_launch_sgd_copy(float* data, const float* delta, float* out,
 void mv_launch_momentum_copy(float* data, float* m, const float* delta,
                              float* out, float mu, int64_t n, hipStream_t s) {
   int64_t n4 = n / 4;
-  if (n4) k_momentum_copy_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
+  if (n4) k_momentum_copy_f4<<<grid_copy(n4), COPY_BLOCK, 0, s>>>(
       (v4f*)data, (v4f*)m, (const v4f*)delta, (v4f*)out, mu, n4);
   int64_t tail = n - n4 * 4;
   if (tail) k_momentum_copy_tail<<<1, 64, 0, s>>>(data, m, delta, out, mu,
@@ -548,7 +560,7 @@ void mv_launch_adagrad_copy(float* data, float* gsq, const float* delta,
                             int64_t n, hipStream_t s) {
   float inv_lr = 1.0f / lr;
   int64_t n4 = n / 4;
-  if (n4) k_adagrad_copy_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
+  if (n4) k_adagrad_copy_f4<<<grid_copy(n4), COPY_BLOCK, 0, s>>>(
       (v4f*)data, (v4f*)gsq, (const v4f*)delta, (v4f*)out, inv_lr, rho, eps,
       n4);
   int64_t tail = n - n4 * 4;
@@ -560,7 +572,7 @@ void mv_launch_dcasgd_copy(float* data, float* bak, const float* delta,
                            float* out, float lr, float lambda, int64_t n,
                            hipStream_t s) {
   int64_t n4 = n / 4;
-  if (n4) k_dcasgd_copy_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
+  if (n4) k_dcasgd_copy_f4<<<grid_copy(n4), COPY_BLOCK, 0, s>>>(
       (v4f*)data, (v4f*)bak, (const v4f*)delta, (v4f*)out, lr, lambda, n4);
   int64_t tail = n - n4 * 4;
   if (tail) k_dcasgd_copy_tail<<<1, 64, 0, s>>>(data, bak, delta, out, lr,
@@ -572,7 +584,7 @@ void mv_launch_dcasgda_copy(float* data, float* bak, float* msq,
                             float lambda, float rho, float eps, int64_t n,
                             hipStream_t s) {
   int64_t n4 = n / 4;
-  if (n4) k_dcasgda_copy_f4<<<grid_for_cap(n4, STATE_GRID), BLOCK, 0, s>>>(
+  if (n4) k_dcasgda_copy_f4<<<grid_copy(n4), COPY_BLOCK, 0, s>>>(
       (v4f*)data, (v4f*)bak, (v4f*)msq, (const v4f*)delta, (v4f*)out, lr,
       lambda, rho, eps, n4);
   int64_t tail = n - n4 * 4;
