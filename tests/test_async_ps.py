@@ -343,3 +343,7 @@ def _reinit_cycle(rank, world):
 
 def test_async_reinit_cycle():
     run_dist(_reinit_cycle, 2)
+
+
+def test_async_soak_ws8():
+    run_dist(_soak, 8, timeout=300)
