@@ -219,27 +219,40 @@ class Zoo:
         return rest
 
     def stop(self, finalize_net: bool = True) -> None:
-        """MV_ShutDown equivalent (zoo.cpp:104-161)."""
+        """MV_ShutDown equivalent (zoo.cpp:104-161). Best-effort after a
+        FatalError: log.fatal tears the process groups down to unblock
+        peers, so every dist call here may find them gone — shutdown
+        must still leave the process exitable."""
         if not self.started:
             return
-        if self.async_engine is not None:
-            eng = self.async_engine
-            eng.shutdown()     # drain + barrier + FinishTrain + join
+        try:
+            if self.async_engine is not None:
+                eng = self.async_engine
+                eng.shutdown()     # drain + barrier + FinishTrain + join
+                self.async_engine = None
+                if dist.is_initialized():
+                    dist.destroy_process_group(eng.req)
+                    dist.destroy_process_group(eng.rep)
+            else:
+                self.barrier()
+        except Exception as e:
+            log.error(f"shutdown after comm teardown (best-effort): {e}")
             self.async_engine = None
-            if dist.is_initialized():
-                dist.destroy_process_group(eng.req)
-                dist.destroy_process_group(eng.rep)
-        else:
-            self.barrier()
         from .tables.base import free_tables
         free_tables()
         self._tables.clear()
         self._next_table_id = 0
-        if dist.is_initialized() and self.control_pg is not None:
-            dist.destroy_process_group(self.control_pg)
+        try:
+            if dist.is_initialized() and self.control_pg is not None:
+                dist.destroy_process_group(self.control_pg)
+        except Exception:
+            pass
         self.control_pg = None
-        if self._owns_pg and dist.is_initialized() and finalize_net:
-            dist.destroy_process_group()
+        try:
+            if self._owns_pg and dist.is_initialized() and finalize_net:
+                dist.destroy_process_group()
+        except Exception:
+            pass
         self.started = False
 
     # ---- bookkeeping (zoo.h:19-85) ----

@@ -347,3 +347,19 @@ def test_async_reinit_cycle():
 
 def test_async_soak_ws8():
     run_dist(_soak, 8, timeout=300)
+
+
+def _sparse_requires_sync(rank, world):
+    """SparseMatrixTable's stale-aware protocol is collective; in async
+    mode construction must refuse LOUDLY (a silent wrong answer would
+    be worse)."""
+    import pytest
+    import multiverso_amd as mv
+    mv.init()
+    with pytest.raises(mv.FatalError, match="sync mode"):
+        mv.SparseMatrixTable(8, 2)
+    mv.shutdown()
+
+
+def test_sparse_matrix_requires_sync_mode():
+    run_dist(_sparse_requires_sync, 2)
