@@ -22,7 +22,13 @@ class LogRegConfig:
     regular_coef: float = 0.0001
     objective_type: str = "sigmoid"     # sigmoid | softmax | ftrl
     updater_type: str = "sgd"           # default | sgd | ftrl | adagrad
-    sparse: bool = False
+    # reference configure.h:24-25 flag: True = libsvm "label k:v ..."
+    # text + sparse model file; False = DENSE "label v v ..." text (the
+    # reference's own mnist.config runs dense softmax) + dense model
+    # file. Default True here (sparse is this framework's primary
+    # path); the reference defaulted false — config files state it
+    # explicitly either way, as the reference's example does.
+    sparse: bool = True
     use_ps: bool = False
     # reference default true (configure.h:84): its effect — overlapping
     # the next chunk's model pull and data parse with training — is

@@ -38,7 +38,7 @@ class LogReg:
         for epoch in range(cfg.train_epoch):
             it = batches if batches is not None else SampleReader(
                 cfg.train_file, cfg.minibatch_size, cfg.reader_type,
-                input_size=cfg.input_size).batches()
+                input_size=cfg.input_size, sparse=cfg.sparse).batches()
             t0 = time.perf_counter()
             nsamples = 0
             losses: List[float] = []
@@ -99,7 +99,7 @@ class LogReg:
         cfg = self.cfg
         it = batches if batches is not None else SampleReader(
             cfg.test_file, cfg.minibatch_size, cfg.reader_type,
-            input_size=cfg.input_size).batches()
+            input_size=cfg.input_size, sparse=cfg.sparse).batches()
         correct = total = 0
         loss_sum = 0.0
         nb = 0
@@ -118,10 +118,16 @@ class LogReg:
                 if float(have[0]) == 0.0:
                     break
                 if b is None:
-                    e = torch.empty(0, dtype=torch.int64)
-                    b = Batch(e, torch.empty(0),
-                              torch.zeros(1, dtype=torch.int64),
-                              torch.empty(0))
+                    if cfg.sparse:
+                        e = torch.empty(0, dtype=torch.int64)
+                        b = Batch(e, torch.empty(0),
+                                  torch.zeros(1, dtype=torch.int64),
+                                  torch.empty(0))
+                    else:
+                        from .objective import DenseBatch
+                        b = DenseBatch(
+                            torch.empty(0, cfg.input_size),
+                            torch.empty(0))
             elif b is None:
                 break
             p = self.model.predict(b)

@@ -26,7 +26,7 @@ import torch
 import multiverso_amd as mv
 from multiverso_amd import ops
 
-from .objective import Batch, create_objective
+from .objective import Batch, DenseBatch, create_objective
 
 
 class WorkerSGDSchedule:
@@ -54,8 +54,14 @@ class LocalModel:
         self.weight = torch.zeros(cfg.input_size, cols, device=self.device)
         self.sched = WorkerSGDSchedule(cfg)
 
-    def update(self, batch: Batch) -> float:
+    def update(self, batch) -> float:
         batch = batch.to(self.device)
+        if isinstance(batch, DenseBatch):
+            # dense data (reference sparse=false): whole-matrix GEMM
+            # gradient on the matrix cores via rocBLAS
+            grad, loss = self.objective.gradient(batch, self.weight)
+            self.weight -= self.sched.next_lr() * grad
+            return loss
         w_rows = self.weight[batch.keys]
         grad, loss = self.objective.gradient(batch, w_rows)
         if self.cfg.objective_type == "ftrl":
@@ -66,8 +72,10 @@ class LocalModel:
             ops.scatter_add_rows(self.weight, batch.keys, grad, -lr)
         return loss
 
-    def predict(self, batch: Batch) -> torch.Tensor:
+    def predict(self, batch) -> torch.Tensor:
         batch = batch.to(self.device)
+        if isinstance(batch, DenseBatch):
+            return self.objective.predict(batch, self.weight)
         return self.objective.predict(batch, self.weight[batch.keys])
 
     # ---- model file io (model.cpp:146-204 / sparse_table.h:258-285) ----
@@ -120,6 +128,8 @@ class PSModel:
         """One sync group: pull union keys, run the minibatches against
         the pulled snapshot (within-chunk updates visible locally), push
         the summed delta (ps_model.cpp:172-203)."""
+        if not self.cfg.sparse:
+            return self._train_chunk_dense(batches)
         batches = [b.to(self.device) for b in batches]
         if batches:
             cat_keys = torch.cat([b.keys for b in batches])
@@ -170,6 +180,34 @@ class PSModel:
         else:
             # server updater 'sgd': w -= delta; push accumulated movement
             self.table.add_rows(union, pulled - local, assume_unique=True)
+        return total_loss / max(len(batches), 1)
+
+    def _train_chunk_dense(self, batches) -> float:
+        """Dense chunk (reference ``sparse=false`` + use_ps — its own
+        mnist.config deployment): the model is pulled WHOLE (all-gather
+        of HBM shards), each minibatch is two GEMMs (scores = X@W,
+        grad = X^T@diff) on the matrix cores via rocBLAS, and the
+        summed movement pushes back as ONE whole-table add
+        (reduce-scatter + updater kernel) — the bandwidth plane end to
+        end. Empty chunks still participate collectively."""
+        from multiverso_amd.log import CHECK
+        batches = [b.to(self.device) for b in batches]
+        pulled = self.table.get()
+        local = pulled.clone()
+        total_loss = 0.0
+        for b in batches:
+            CHECK(isinstance(b, DenseBatch),
+                  "cfg.sparse=false requires dense batches "
+                  "(\"label value value ...\" data)")
+            grad, loss = self.objective.gradient(b, local)
+            total_loss += loss
+            local -= self.sched.next_lr() * grad
+        if self.table.updater_type == "adagrad":
+            opt = mv.AddOption(learning_rate=1.0,
+                               rho=self.cfg.learning_rate)
+            self.table.add(pulled - local, option=opt).wait()
+        else:
+            self.table.add(pulled - local).wait()
         return total_loss / max(len(batches), 1)
 
     def _fused_kind(self, local: torch.Tensor) -> str:
@@ -240,8 +278,12 @@ class PSModel:
             loss_acc += lossb.mean()
         return float(loss_acc)
 
-    def predict(self, batch: Batch) -> torch.Tensor:
+    def predict(self, batch) -> torch.Tensor:
         batch = batch.to(self.device)
+        if isinstance(batch, DenseBatch):
+            # whole-model pull (collective — dense ranks' empty
+            # participation batches are DenseBatches too)
+            return self.objective.predict(batch, self.table.get())
         w_rows = self.table.get_rows(batch.keys)
         return self.objective.predict(batch, w_rows)
 

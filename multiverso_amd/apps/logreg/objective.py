@@ -21,6 +21,30 @@ import torch
 from multiverso_amd import ops
 
 
+class DenseBatch:
+    """Dense minibatch (reference ``sparse=false`` data,
+    configure.h:60-63: "label value value ..."): ``x`` is [B, d]
+    float32 with the bias column (value 1) LAST — the reference appends
+    the bias value to every sample (reader.cpp:235-236)."""
+
+    __slots__ = ("x", "labels", "weights")
+
+    def __init__(self, x: torch.Tensor, labels: torch.Tensor,
+                 weights: Optional[torch.Tensor] = None) -> None:
+        self.x = x
+        self.labels = labels
+        self.weights = weights
+
+    @property
+    def size(self) -> int:
+        return self.x.size(0)
+
+    def to(self, device) -> "DenseBatch":
+        return DenseBatch(self.x.to(device), self.labels.to(device),
+                          None if self.weights is None
+                          else self.weights.to(device))
+
+
 class Batch:
     """Sparse minibatch: sample i has keys[ptr[i]:ptr[i+1]] etc."""
 
@@ -51,13 +75,22 @@ class Batch:
             torch.arange(self.size, device=self.keys.device), lens)
 
 
-def _scores(batch: Batch, w_rows: torch.Tensor) -> torch.Tensor:
+def _scores(batch, w_rows: torch.Tensor) -> torch.Tensor:
     """scores[B, O] = sum_j x_j * W[key_j, :] per sample (K13).
 
-    The batch is CSR (ptr segments a sample's features contiguously), so
-    the per-sample sum is a segmented reduction: cumsum + boundary diff —
-    no atomics (measured 3x the throughput of the scatter-add form on
-    gfx950; fp64 accumulator keeps the diff exact for fp32 data)."""
+    Dense batches: one GEMM, X[B,d] @ W[d,O] — routed through
+    torch.matmul = rocBLAS on ROCm, i.e. the MFMA matrix cores (the
+    library is the right owner of compute-bound plain GEMMs; see
+    docs/ENGINEERING_NOTES.md "MFMA decision"). ``w_rows`` is then the
+    FULL weight matrix.
+
+    Sparse batches are CSR (ptr segments a sample's features
+    contiguously), so the per-sample sum is a segmented reduction:
+    cumsum + boundary diff — no atomics (measured 3x the throughput of
+    the scatter-add form on gfx950; fp64 accumulator keeps the diff
+    exact for fp32 data)."""
+    if isinstance(batch, DenseBatch):
+        return batch.x @ w_rows
     contrib = (batch.vals.unsqueeze(1) * w_rows).to(torch.float64)
     cs = torch.empty(contrib.size(0) + 1, contrib.size(1),
                      device=contrib.device, dtype=torch.float64)
@@ -115,12 +148,19 @@ class Objective:
     def predict(self, batch: Batch, w_rows: torch.Tensor) -> torch.Tensor:
         raise NotImplementedError
 
-    def gradient(self, batch: Batch, w_rows: torch.Tensor
+    def gradient(self, batch, w_rows: torch.Tensor
                  ) -> Tuple[torch.Tensor, float]:
         p = self.predict(batch, w_rows)
         diff = p - _one_hot(batch.labels, self.output_size)
         if batch.weights is not None:
             diff = diff * batch.weights.unsqueeze(1)
+        if isinstance(batch, DenseBatch):
+            # dense: grad[d, O] = X^T @ diff — the second GEMM of the
+            # pair (also rocBLAS/MFMA); regularizer applies to the full
+            # weight matrix (reference AddRegularization over the whole
+            # dense model)
+            grad = batch.x.t() @ diff + self.regular.delta(w_rows)
+            return grad, self.loss(batch, p)
         grad = batch.vals.unsqueeze(1) * diff[batch.sample_ids()]
         grad = grad + self.regular.delta(w_rows)
         loss = self.loss(batch, p)
@@ -185,8 +225,15 @@ class FTRLObjective(Objective):
     def predict(self, batch: Batch, zn_rows: torch.Tensor) -> torch.Tensor:
         return torch.sigmoid(_scores(batch, self.reconstruct_w(zn_rows)))
 
-    def gradient(self, batch: Batch, zn_rows: torch.Tensor
+    def gradient(self, batch, zn_rows: torch.Tensor
                  ) -> Tuple[torch.Tensor, float]:
+        if isinstance(batch, DenseBatch):
+            # FTRL's per-occurrence z/n state math is sparse by
+            # construction (the reference pairs FTRL with its sparse
+            # tables, ftrl_sparse_table.h); refuse dense loudly.
+            from multiverso_amd.log import CHECK
+            CHECK(False, "FTRL objective requires sparse data "
+                         "(objective_type=ftrl with sparse=false)")
         O = self.output_size
         w = self.reconstruct_w(zn_rows)
         p = torch.sigmoid(_scores(batch, w))

@@ -224,3 +224,109 @@ def test_logreg_uneven_files_dist(tmp_path):
     import functools
     from conftest import run_dist
     run_dist(functools.partial(_logreg_uneven_files, tmpdir=str(tmp_path)), 2)
+
+
+# ---- dense data mode (reference sparse=false, its own mnist.config) ----
+
+def test_dense_reader_parse(tmp_path):
+    from multiverso_amd.apps.logreg.objective import DenseBatch
+    p = tmp_path / "dense.txt"
+    p.write_text("1 0.5 2.0 -1.0\n0 1.0 0.0 3.0\n")
+    b = list(SampleReader(str(p), 2, input_size=4,
+                          sparse=False).batches())[0]
+    assert isinstance(b, DenseBatch)
+    assert b.x.shape == (2, 4)       # 3 features + bias
+    assert torch.equal(b.x[0], torch.tensor([0.5, 2.0, -1.0, 1.0]))
+    assert torch.equal(b.labels, torch.tensor([1.0, 0.0]))
+
+
+def _synthetic_dense(n, d, K, seed=0):
+    """Separable dense stream: x ~ N(0,1), label = argmax(x @ hidden)."""
+    from multiverso_amd.apps.logreg.objective import DenseBatch
+    g = torch.Generator().manual_seed(seed)
+    hidden = torch.randn(d, K, generator=g)
+    x = torch.randn(n, d, generator=g)
+    scores = x @ hidden
+    labels = (scores.argmax(1).float() if K > 1
+              else (scores.squeeze(1) > 0).float())
+    x = torch.cat([x, torch.ones(n, 1)], dim=1)   # bias column
+    return x, labels
+
+
+def test_dense_local_softmax_learns(env):
+    from multiverso_amd.apps.logreg.objective import DenseBatch
+    cfg = LogRegConfig(input_size=33, output_size=4,
+                       objective_type="softmax", sparse=False,
+                       minibatch_size=32, learning_rate=0.2,
+                       learning_rate_coef=1e6, show_time_per_sample=0)
+    lr = LogReg(cfg)
+    x, labels = _synthetic_dense(960, 32, 4, seed=3)
+    batches = [DenseBatch(x[i:i + 32], labels[i:i + 32])
+               for i in range(0, 960, 32)]
+    lr.train(iter(batches))
+    acc, _ = lr.test(iter(batches[:10]))
+    assert acc > 0.8, acc
+
+
+def _dense_ps_dist(rank, world):
+    """Dense PS mode at ws2: whole-table pull/push on the collective
+    plane; per-rank test sets of DIFFERENT sizes exercise the dense
+    empty-batch participation."""
+    import torch
+    import multiverso_amd as mv
+    from multiverso_amd.apps.logreg import LogReg, LogRegConfig
+    from multiverso_amd.apps.logreg.objective import DenseBatch
+    from test_logreg import _synthetic_dense
+    mv.init(sync=True)
+    cfg = LogRegConfig(input_size=17, output_size=3,
+                       objective_type="softmax", sparse=False,
+                       use_ps=True, sync_frequency=2, minibatch_size=16,
+                       learning_rate=0.2, learning_rate_coef=1e6,
+                       show_time_per_sample=0)
+    # SAME hidden label function on every rank (seed fixed); each rank
+    # trains its own interleaved sample shard
+    x, labels = _synthetic_dense(640, 16, 3, seed=11)
+    x, labels = x[rank::world], labels[rank::world]
+    batches = [DenseBatch(x[i:i + 16], labels[i:i + 16])
+               for i in range(0, 320, 16)]
+    lr = LogReg(cfg)
+    lr.train(iter(batches))
+    n_test = 6 if rank == 0 else 3       # uneven participation
+    acc, _ = lr.test(iter(batches[:n_test]))
+    assert acc > 0.7, (rank, acc)
+    mv.shutdown()
+
+
+def test_dense_ps_dist():
+    run_dist(_dense_ps_dist, 2)
+
+
+def test_dense_cli_end_to_end(tmp_path):
+    """Reference-style dense deployment: a sparse=false config file +
+    'label value value ...' text through the CLI main."""
+    import subprocess, sys, os
+    d, K = 8, 3
+    x, labels = _synthetic_dense(240, d, K, seed=9)
+    train = tmp_path / "train.data"
+    with open(train, "w") as f:
+        for i in range(240):
+            vals = " ".join(f"{v:.4f}" for v in x[i, :d].tolist())
+            f.write(f"{int(labels[i])} {vals}\n")
+    cfgf = tmp_path / "dense.config"
+    cfgf.write_text(
+        f"input_size={d + 1}\noutput_size={K}\nobjective_type=softmax\n"
+        "sparse=false\ntrain_epoch=6\nminibatch_size=24\n"
+        "learning_rate=0.2\nlearning_rate_coef=1000000\n"
+        f"train_file={train}\ntest_file={train}\n"
+        "show_time_per_sample=0\n")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env2 = dict(os.environ)
+    env2["PYTHONPATH"] = repo + ":" + env2.get("PYTHONPATH", "")
+    r = subprocess.run(
+        [sys.executable, "-m", "multiverso_amd.apps.logreg.main",
+         str(cfgf)], capture_output=True, text=True, timeout=300,
+        env=env2)
+    assert r.returncode == 0, r.stderr[-2000:]
+    import re
+    m = re.findall(r"correct \(([0-9.]+)\)", r.stdout + r.stderr)
+    assert m and float(m[-1]) > 0.8, (m, r.stdout[-800:])
