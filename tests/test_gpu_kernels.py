@@ -646,3 +646,31 @@ def test_w2v_dim_above_limit_refuses_loudly():
     with pytest.raises(RuntimeError, match="2048"):
         hip.w2v_train_ns(emb, emb, emb, emb, ids, None, ids, pool,
                          2, 1, 0.05, False, True)
+
+
+def test_dense_logreg_gpu():
+    """Dense data mode on GPU: the GEMM pair (X@W, X^T@diff) runs on
+    device via rocBLAS (matrix cores) and learning works."""
+    import multiverso_amd as mv
+    from multiverso_amd.apps.logreg import LogReg, LogRegConfig
+    from multiverso_amd.apps.logreg.objective import DenseBatch
+    mv.init(sync=True)
+    torch.manual_seed(5)
+    d, K, n = 256, 8, 2048
+    hidden = torch.randn(d, K, device="cuda:0")
+    x = torch.randn(n, d, device="cuda:0")
+    labels = (x @ hidden).argmax(1).float()
+    x = torch.cat([x, torch.ones(n, 1, device="cuda:0")], dim=1)
+    cfg = LogRegConfig(input_size=d + 1, output_size=K,
+                       objective_type="softmax", sparse=False, use_ps=True,
+                       sync_frequency=2, minibatch_size=128,
+                       learning_rate=0.2, learning_rate_coef=1e6,
+                       show_time_per_sample=0)
+    batches = [DenseBatch(x[i:i + 128], labels[i:i + 128])
+               for i in range(0, n, 128)]
+    lr = LogReg(cfg)
+    lr.train(iter(batches))
+    acc, _ = lr.test(iter(batches[:4]))
+    assert acc > 0.8, acc
+    assert lr.model.table.shard.is_cuda
+    mv.shutdown()
