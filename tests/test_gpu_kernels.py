@@ -669,7 +669,7 @@ def test_dense_logreg_gpu():
     batches = [DenseBatch(x[i:i + 128], labels[i:i + 128])
                for i in range(0, n, 128)]
     lr = LogReg(cfg)
-    lr.train(iter(batches))
+    lr.train(iter(batches * 5))      # 5 passes: 256-dim 8-class needs them
     acc, _ = lr.test(iter(batches[:4]))
     assert acc > 0.8, acc
     assert lr.model.table.shard.is_cuda
