@@ -52,6 +52,9 @@ def parse_args():
                    help="logreg: objective (fused kernel per objective)")
     p.add_argument("--classes", type=int, default=10,
                    help="logreg softmax: number of classes")
+    p.add_argument("--dense", action="store_true",
+                   help="logreg: dense data mode (reference sparse=false "
+                        "— GEMMs on the matrix cores via rocBLAS)")
     return p.parse_args()
 
 

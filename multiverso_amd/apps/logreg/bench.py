@@ -32,9 +32,18 @@ def run_bench(args):
 
     objective = getattr(args, "objective", "sigmoid")
     classes = getattr(args, "classes", 10) if objective == "softmax" else 1
+    dense = bool(getattr(args, "dense", False))
+    if dense:
+        # reference mnist.config shape class: dense softmax; input_size
+        # is the dense width (incl. bias), far below the sparse 1e9
+        input_size = getattr(args, "features", 2048)
+        if input_size > 1_000_000:
+            input_size = 2048
+        if not cuda:
+            input_size = min(input_size, 256)
     cfg = LogRegConfig(input_size=input_size, minibatch_size=minibatch,
-                       use_ps=True,
-                       updater_type="sgd" if objective == "ftrl"
+                       use_ps=True, sparse=not dense,
+                       updater_type="sgd" if (objective == "ftrl" or dense)
                        else "adagrad",
                        objective_type=objective, output_size=classes,
                        sync_frequency=sync_freq, learning_rate=0.05,
@@ -42,9 +51,24 @@ def run_bench(args):
     model = PSModel(cfg, device)
 
     n_chunks = args.warmup + args.steps
-    batches, _ = synthetic_batches(input_size, n_chunks * sync_freq,
-                                   minibatch, nnz=nnz, output_size=classes,
-                                   seed=31 + rank, device=device)
+    if dense:
+        from .objective import DenseBatch
+        g = torch.Generator(device=device).manual_seed(31 + rank)
+        hidden = torch.randn(input_size, classes, generator=g,
+                             device=device)
+        batches = []
+        for _ in range(n_chunks * sync_freq):
+            x = torch.randn(minibatch, input_size, generator=g,
+                            device=device)
+            x[:, -1] = 1.0   # bias column
+            lab = ((x @ hidden).argmax(1).float() if classes > 1
+                   else ((x @ hidden).squeeze(1) > 0).float())
+            batches.append(DenseBatch(x, lab))
+    else:
+        batches, _ = synthetic_batches(input_size, n_chunks * sync_freq,
+                                       minibatch, nnz=nnz,
+                                       output_size=classes,
+                                       seed=31 + rank, device=device)
     chunks = [batches[i * sync_freq:(i + 1) * sync_freq]
               for i in range(n_chunks)]
 
@@ -73,7 +97,8 @@ def run_bench(args):
 
     if rank == 0:
         print(json.dumps({
-            "metric": "LogReg sparse samples/sec (whole node)",
+            "metric": ("LogReg dense samples/sec (whole node)" if dense
+                       else "LogReg sparse samples/sec (whole node)"),
             "value": sps,
             "unit": "samples/s",
             "n_gpus": n,
@@ -86,8 +111,11 @@ def run_bench(args):
             "dtype": "fp32",
             "data": "synthetic",
             "config": {
-                "model": f"sparse logistic regression {input_size} features "
-                         f"nnz={nnz} {objective}"
+                "model": (f"dense logistic regression {input_size} "
+                          f"features {objective}"
+                          if dense else
+                          f"sparse logistic regression {input_size} "
+                          f"features nnz={nnz} {objective}")
                          + (f" K={classes}" if classes > 1 else ""),
                 "global_batch": n * minibatch * sync_freq,
                 "seq_len": None,
