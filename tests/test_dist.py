@@ -70,10 +70,15 @@ def _matrix_same_rows(rank, world):
 
 
 def _kv(rank, world):
+    # default mode is TRUE ASYNC: my own adds are visible immediately
+    # (FIFO per pair); everyone's adds only after a drain barrier
     import multiverso_amd as mv
     mv.init()
     t = mv.KVTable()
     t.add([1, 2, 3], [1.0, 2.0, 3.0])
+    got = t.get([1])
+    assert got[1] >= 1.0, got
+    mv.barrier()
     got = t.get([1, 2, 3])
     assert got == {1: 1.0 * world, 2: 2.0 * world, 3: 3.0 * world}
     mv.shutdown()
