@@ -18,7 +18,7 @@ setup(
               + ["multiverso"]),
     package_dir={"multiverso": "binding/python/multiverso"},
     package_data={"multiverso_amd.ops": ["csrc/*", "_build/*.so"],
-                  "multiverso_amd.capi": ["*.h", "*.cpp", "*.so"]},
+                  "multiverso_amd.capi": ["*.h", "*.hpp", "*.cpp", "*.so"]},
     python_requires=">=3.8",
     install_requires=["torch"],
 )
