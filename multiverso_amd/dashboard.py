@@ -78,10 +78,16 @@ class Dashboard:
 
 @contextmanager
 def monitor(name: str):
-    """MONITOR_BEGIN/END equivalent (reference dashboard.h:61-74)."""
+    """MONITOR_BEGIN/END equivalent (reference dashboard.h:61-74).
+
+    Uses a LOCAL timestamp, not the Monitor's embedded timer: the async
+    server thread and the caller's thread time the same monitor names
+    concurrently, and a shared timer would interleave begin/end pairs
+    (the counters themselves are GIL-atomic float adds)."""
     m = Dashboard.get(name)
-    m.begin()
+    t0 = time.perf_counter()
     try:
         yield m
     finally:
-        m.end()
+        m.count += 1
+        m.elapsed_ms += (time.perf_counter() - t0) * 1e3
