@@ -56,6 +56,8 @@ OP_KV_ADD = 5       # KV: int64 keys + f64 value deltas
 OP_KV_GET = 6       # KV: int64 keys; reply = f64 values
 OP_FLUSH = 7        # fence: ack after all my prior requests are applied
 OP_FINISH = 8       # worker is done (Zoo StopPS / FinishTrain parity)
+OP_GET_STALE = 9    # sparse table: rows stale FOR THIS WORKER; reply =
+                    # count + global ids + values (matrix.cpp:478,540)
 
 _HDR = 6  # int64 fields: [op, table_id, n_keys, n_vals, want_ack, has_opt]
 
@@ -296,6 +298,44 @@ class AsyncEngine:
             out[m] = buf
         return out
 
+    def sparse_get_stale(self, table, cache: torch.Tensor) -> int:
+        """Stale-aware whole-table Get under async (the reference's
+        sparse table ran under its async server the same way,
+        matrix.cpp:461-478): each owner serves the rows stale for THIS
+        worker on arrival and marks them fresh. Reply sizes are unknown
+        to the requester, so the reply is count-then-payload on the FIFO
+        pair (blocking recvs in send order). Returns rows received."""
+        zoo = self.zoo
+        unit = table.num_col
+        me = zoo.worker_id
+        total = 0
+        remotes = []
+        for s in range(table.spec.n):
+            dst = zoo.server_ranks[s]
+            if dst == zoo.rank:
+                ids, vals = table._server_stale_for(me)
+                if ids.numel():
+                    cache[ids] = vals.to(cache.device)
+                    total += ids.numel()
+                continue
+            hdr = [OP_GET_STALE, table.table_id, 0, 0, 0, 0]
+            for w in self._send_request(dst, hdr, []):
+                w.wait()
+            remotes.append(dst)
+        for dst in remotes:
+            cnt = torch.empty(1, dtype=torch.int64)
+            dist.recv(cnt, dst, group=self.rep)
+            k = int(cnt[0])
+            if not k:
+                continue
+            ids = torch.empty(k, dtype=torch.int64)
+            dist.recv(ids, dst, group=self.rep)
+            vals = torch.empty(k * unit, dtype=table.dtype)
+            dist.recv(vals, dst, group=self.rep)
+            cache[ids] = vals.view(k, unit).to(cache.device)
+            total += k
+        return total
+
     # ---- fences ----
     def drain(self) -> None:
         """Fence: after this, every request THIS worker issued has been
@@ -431,5 +471,15 @@ class AsyncEngine:
             table._server_kv_add(keys, vals)
         elif op == OP_KV_GET:
             dist.send(table._server_kv_get(keys), src, group=self.rep)
+        elif op == OP_GET_STALE:
+            worker = self.zoo.worker_ranks.index(src)
+            with monitor("server.process_get"):
+                ids, served = table._server_stale_for(worker)
+            dist.send(torch.tensor([ids.numel()], dtype=torch.int64), src,
+                      group=self.rep)
+            if ids.numel():
+                dist.send(ids.cpu().contiguous(), src, group=self.rep)
+                dist.send(served.reshape(-1).cpu().contiguous(), src,
+                          group=self.rep)
         else:
             CHECK(False, f"unknown async op {op}")

@@ -68,7 +68,10 @@ class MatrixTable(Table):
         # headline config). Any other table op materializes the Add
         # first (flush).
         self._deferred = None  # (delta, option, delta._version)
-        self._ready.set()
+        # subclasses with extra server state (SparseMatrixTable's
+        # bitmap) publish readiness themselves after that state exists
+        if not getattr(type(self), "_defer_ready", False):
+            self._ready.set()
 
     def _check_deferred(self, d) -> None:
         CHECK(d[0]._version == d[2],

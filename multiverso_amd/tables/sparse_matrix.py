@@ -30,19 +30,13 @@ from .matrix_table import MatrixTable
 
 
 class SparseMatrixTable(MatrixTable):
+    _defer_ready = True   # bitmap must exist before async requests serve
+
     def __init__(self, num_row: int, num_col: int,
                  dtype: torch.dtype = torch.float32,
                  updater_type: Optional[str] = None,
                  random_init=None) -> None:
         super().__init__(num_row, num_col, dtype, updater_type, random_init)
-        # The stale-aware whole-table Get is a coordinated protocol
-        # (every owner builds every requester's stale list in lockstep,
-        # sparse_matrix_table.cpp:226-258); it runs on the BSP collective
-        # plane. Async deployments use MatrixTable row ops instead.
-        CHECK(self.engine is None,
-              "SparseMatrixTable requires sync mode (-sync=true): its "
-              "stale-row exchange is collective; async-mode training "
-              "should use MatrixTable.get_rows/add_rows")
         # SparseFilter on the stale-row reply payloads (the reference
         # filters every outgoing sparse-table bundle,
         # sparse_matrix_table.cpp:148-153): per-destination segments are
@@ -56,6 +50,7 @@ class SparseMatrixTable(MatrixTable):
         # up_to_date[w, r] == True -> worker w has the current row r
         self.up_to_date = torch.zeros(nw, self.local_rows, dtype=torch.bool,
                                       device=self.device)
+        self._ready.set()
 
     # ---- invalidation hooks ----
     def _invalidate_all_but(self, local_ids: torch.Tensor,
@@ -71,9 +66,46 @@ class SparseMatrixTable(MatrixTable):
         self.up_to_date.zero_()
         return h
 
+    # ---- async-mode server hooks (the reference's sparse table served
+    # its bitmap under the async server identically, matrix.cpp:461-478) --
+    def _server_apply_chunk(self, chunk, option) -> None:
+        super()._server_apply_chunk(chunk, option)
+        with self._shard_lock:
+            self.up_to_date.zero_()   # whole-shard add: everyone stale
+
+    def _server_stale_for(self, worker: int):
+        """Rows of MY shard stale for ``worker``: gather them, mark them
+        fresh, return (global_ids, values). UpdateGetState semantics
+        (sparse_matrix_table.cpp:226-258) served on arrival."""
+        with self._shard_lock:
+            stale = (~self.up_to_date[worker]).nonzero().reshape(-1)
+            if stale.numel():
+                served = self._gather_local(stale)
+                self.up_to_date[worker, stale] = True
+            else:
+                served = torch.empty(0, self.num_col, dtype=self.dtype,
+                                     device=self.device)
+            return stale + self.row_offset, served
+
+    def _server_add_rows(self, local_ids, vals2d, option) -> None:
+        super()._server_add_rows(local_ids, vals2d, option)
+        with self._shard_lock:
+            self.up_to_date[:, local_ids.to(self.up_to_date.device)] = False
+
     def add_rows(self, row_ids, values, option: Optional[AddOption] = None,
                  source_worker: Optional[int] = None) -> None:
         self.flush()   # a deferred whole-table Add must land first
+        eng = self.engine
+        if eng is not None:
+            # keyed add served on arrival; the server-side hook above
+            # invalidates the touched rows for every worker
+            ids = torch.as_tensor(row_ids, dtype=torch.int64).cpu()
+            vals = values.to(self.device, self.dtype).contiguous()
+            from ..comm import Handle
+            with monitor("worker.add_rows"):
+                self._track(Handle(eng.keyed_add(
+                    self, ids, vals, self.num_col, option)))
+            return
         ids = torch.as_tensor(row_ids, dtype=torch.int64)
         vals = values.to(self.device, self.dtype).contiguous()
         from ..comm import all_to_all_rows
@@ -143,6 +175,11 @@ class SparseMatrixTable(MatrixTable):
         CHECK(cache.shape == (self.num_row, self.num_col),
               "cache must be the full table shape")
         self.flush()
+        if self.engine is not None:
+            n = self.get_into(cache)   # async lane: served on arrival
+            h = Handle()
+            h.rows = n
+            return h
         if self.zoo.size == 1:
             n = self.get_into(cache)
             h = Handle()
@@ -192,6 +229,11 @@ class SparseMatrixTable(MatrixTable):
         CHECK(cache.shape == (self.num_row, self.num_col),
               "cache must be the full table shape")
         self.flush()
+        eng = self.engine
+        if eng is not None:
+            CHECK(self.zoo.is_worker, "ps_role=server ranks issue no Gets")
+            with monitor("worker.sparse_get"):
+                return eng.sparse_get_stale(self, cache)
         me = self.zoo.worker_id
         nw = self.zoo.num_workers
         with monitor("worker.sparse_get"):

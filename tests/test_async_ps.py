@@ -350,17 +350,46 @@ def test_async_soak_ws8():
     run_dist(_soak, 8, timeout=300)
 
 
-def _sparse_requires_sync(rank, world):
-    """SparseMatrixTable's stale-aware protocol is collective; in async
-    mode construction must refuse LOUDLY (a silent wrong answer would
-    be worse)."""
-    import pytest
+# (round-2 late: SparseMatrixTable now WORKS under async — see
+# test_sparse_async below — so the former construction refusal is gone)
+
+
+def _sparse_async(rank, world):
+    """SparseMatrixTable under TRUE ASYNC (the reference's sparse table
+    ran under its async server, matrix.cpp:461-478): stale-row gets are
+    served on arrival, ranks may call get_into UNEQUAL numbers of times
+    (impossible on the collective plane), adds invalidate for every
+    worker."""
     import multiverso_amd as mv
     mv.init()
-    with pytest.raises(mv.FatalError, match="sync mode"):
-        mv.SparseMatrixTable(8, 2)
+    t = mv.SparseMatrixTable(9, 3)
+    cache = torch.zeros(9, 3)
+    n = t.get_into(cache)            # initial full pull
+    assert n == 9, n
+    assert t.get_into(cache) == 0    # fresh now
+    if rank == 0:
+        t.add_rows([1, 8], torch.ones(2, 3))
+        # extra gets only on rank 0 — a collective plane would hang
+        t.get_into(cache)
+        t.get_into(cache)
+    mv.barrier()                     # rank 0's adds visible everywhere
+    n = t.get_into(cache)
+    if rank != 0:
+        assert n == 2, (rank, n)
+        assert torch.equal(cache[1], torch.ones(3))
+        assert torch.equal(cache[8], torch.ones(3))
+    mv.barrier()
+    # whole-table add invalidates every worker's rows on every shard
+    t.add(torch.full((9, 3), 0.5))
+    mv.barrier()
+    n = t.get_into(cache)
+    assert n == 9, (rank, n)
     mv.shutdown()
 
 
-def test_sparse_matrix_requires_sync_mode():
-    run_dist(_sparse_requires_sync, 2)
+def test_sparse_async():
+    run_dist(_sparse_async, 2)
+
+
+def test_sparse_async_ws3():
+    run_dist(_sparse_async, 3)
