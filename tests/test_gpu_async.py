@@ -43,6 +43,17 @@ def _async_gpu_shards(rank, world):
     expect[1] = expect[2] = 1.0      # 63-0=63, 63-1=62
     assert torch.equal(rows.cpu(), expect), (rank, rows)
 
+    # sparse stale service with HBM shards (OP_GET_STALE)
+    sp = mv.SparseMatrixTable(32, 4)
+    cache = torch.zeros(32, 4)
+    assert sp.get_into(cache) == 32
+    assert sp.get_into(cache) == 0
+    sp.add_rows([rank, 31 - rank], torch.ones(2, 4))
+    mv.barrier()
+    n = sp.get_into(cache)
+    assert n == 4, (rank, n)     # both ranks' adds stale for me
+    assert torch.equal(cache[0], torch.ones(4))
+
     # unequal op counts with GPU shards (the async signature move)
     t3 = mv.ArrayTable(128)
     for _ in range(2 if rank == 0 else 5):
