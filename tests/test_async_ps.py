@@ -367,6 +367,10 @@ def _sparse_async(rank, world):
     n = t.get_into(cache)            # initial full pull
     assert n == 9, n
     assert t.get_into(cache) == 0    # fresh now
+    # fence BEFORE rank 0 adds: without it a fast rank 0's add can
+    # invalidate rows before a slow rank's freshness assert above —
+    # correct async behavior, wrong test oracle
+    mv.barrier()
     if rank == 0:
         t.add_rows([1, 8], torch.ones(2, 3))
         # extra gets only on rank 0 — a collective plane would hang
