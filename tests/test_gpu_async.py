@@ -48,6 +48,7 @@ def _async_gpu_shards(rank, world):
     cache = torch.zeros(32, 4)
     assert sp.get_into(cache) == 32
     assert sp.get_into(cache) == 0
+    mv.barrier()   # fence: a fast peer's add must not precede the ==0
     sp.add_rows([rank, 31 - rank], torch.ones(2, 4))
     mv.barrier()
     n = sp.get_into(cache)
