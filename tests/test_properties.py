@@ -71,3 +71,45 @@ def test_huffman_invariants(counts):
         assert len(l.point) == len(l.code) >= 1
         assert all(0 <= p < v - 1 for p in l.point)
         assert l.point[0] == v - 2  # root first (reference layout)
+
+
+@given(total=st.integers(min_value=1, max_value=100_000),
+       n=st.integers(min_value=1, max_value=16),
+       seed=st.integers(min_value=0, max_value=1_000))
+@settings(max_examples=60, deadline=None)
+def test_keyed_partition_matches_shard_owner(total, n, seed):
+    """The async engine's keyed plan and the collective plane's owner
+    math must agree with ShardSpec.range_of for EVERY id: the id lands
+    on the server whose [offset, offset+count) range contains it."""
+    import torch
+    from multiverso_amd.comm import ShardSpec
+    spec = ShardSpec(total, n)
+    g = torch.Generator().manual_seed(seed)
+    ids = torch.randint(0, total, (min(total, 64),), generator=g)
+    base = max(total // n, 1)
+    owners = torch.div(ids, base, rounding_mode="floor").clamp_(max=n - 1)
+    for i, o in zip(ids.tolist(), owners.tolist()):
+        off, cnt = spec.range_of(o)
+        assert off <= i < off + cnt, (i, o, off, cnt)
+        assert spec.owner_of(i) == o
+
+
+@given(sizes=st.lists(st.integers(min_value=0, max_value=1 << 20),
+                      min_size=1, max_size=8),
+       flags=st.data())
+@settings(max_examples=30, deadline=None)
+def test_addoption_wire_roundtrip(sizes, flags):
+    """AddOption's 20-byte envelope and the engine's f64[5] transport
+    encoding both round-trip exactly."""
+    from multiverso_amd.updaters import AddOption
+    from multiverso_amd.async_ps import _opt_from, _opt_tensor
+    o = AddOption(worker_id=sizes[0] % 1024, momentum=0.25,
+                  learning_rate=0.125, rho=0.5, lambda_=0.0625)
+    o2 = AddOption.from_bytes(o.to_bytes())
+    assert (o2.worker_id, o2.momentum, o2.learning_rate, o2.rho,
+            o2.lambda_) == (o.worker_id, o.momentum, o.learning_rate,
+                            o.rho, o.lambda_)
+    o3 = _opt_from(_opt_tensor(o))
+    assert (o3.worker_id, o3.momentum, o3.learning_rate, o3.rho,
+            o3.lambda_) == (o.worker_id, o.momentum, o.learning_rate,
+                            o.rho, o.lambda_)
