@@ -7,7 +7,10 @@ import subprocess
 import sys
 import os
 
+import pytest
+
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+REPO = ROOT
 
 
 def test_wordembedding_cli(tmp_path):
@@ -171,3 +174,24 @@ def test_embedding_server(tmp_path):
     assert nn2["neighbors"][0]["word"] == "pear"
     assert client.get("/vec/zzz").status_code == 404
     mv.shutdown()
+
+
+@pytest.mark.parametrize("name", ["array", "kv", "net", "matrix",
+                                  "allreduce"])
+def test_selftest_dispatcher_ws2(name, tmp_path):
+    """The reference's `mpirun -np N ./multiverso.test <name>` CLI tests
+    (Test/main.cpp:12-24) as a torchrun dispatcher."""
+    import subprocess, sys, os
+    from conftest import free_port
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + ":" + env.get("PYTHONPATH", "")
+    env["HIP_VISIBLE_DEVICES"] = env["CUDA_VISIBLE_DEVICES"] = ""
+    env["MV_BACKEND"] = "gloo"
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(free_port()),
+         "-m", "multiverso_amd.selftest", name],
+        capture_output=True, text=True, timeout=300, env=env)
+    assert r.returncode == 0, (name, r.stdout[-500:], r.stderr[-1500:])
+    assert "PASS" in r.stdout
