@@ -674,3 +674,19 @@ def test_dense_logreg_gpu():
     assert acc > 0.8, acc
     assert lr.model.table.shard.is_cuda
     mv.shutdown()
+
+
+def test_f64_matrix_keyed_gpu():
+    """float64 MatrixTable keyed ops on GPU: torch-indexing parity path
+    (the HIP keyed kernels are the f32 hot path; f64 must still be
+    correct, not crash into check_f32)."""
+    import multiverso_amd as mv
+    mv.init()
+    t = mv.MatrixTable(64, 4, dtype=torch.float64)
+    t.add_rows([3, 60], torch.ones(2, 4, dtype=torch.float64) * 2.5)
+    got = t.get_rows([3, 60, 5])
+    assert got.dtype == torch.float64 and got.is_cuda
+    assert torch.equal(got.cpu()[0], torch.full((4,), 2.5,
+                                                dtype=torch.float64))
+    assert torch.equal(got.cpu()[2], torch.zeros(4, dtype=torch.float64))
+    mv.shutdown()

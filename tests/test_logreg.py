@@ -330,3 +330,17 @@ def test_dense_cli_end_to_end(tmp_path):
     import re
     m = re.findall(r"correct \(([0-9.]+)\)", r.stdout + r.stderr)
     assert m and float(m[-1]) > 0.8, (m, r.stdout[-800:])
+
+
+def test_dense_weighted_reader_parse(tmp_path):
+    """Weighted dense lines: 'label weight value ...' (reference
+    WeightedSampleReader over dense data)."""
+    from multiverso_amd.apps.logreg.objective import DenseBatch
+    p = tmp_path / "wd.txt"
+    p.write_text("1 0.5 2.0 3.0\n0 2.0 -1.0 1.0\n")
+    b = list(SampleReader(str(p), 2, "weight", input_size=3,
+                          sparse=False).batches())[0]
+    assert isinstance(b, DenseBatch)
+    assert torch.equal(b.x, torch.tensor([[2.0, 3.0, 1.0],
+                                          [-1.0, 1.0, 1.0]]))
+    assert torch.equal(b.weights, torch.tensor([0.5, 2.0]))
