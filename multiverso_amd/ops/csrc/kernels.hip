@@ -32,6 +32,11 @@
 // still: 6.05/5.84 TB/s at 768 blocks vs 5.45/5.15 at 1024 (same sweep).
 #define ELEM_GRID 1024
 #define STATE_GRID 768
+// 3-stream update kernels (read data + read delta + write data) peak at
+// 768 blocks, not 1024: 6.05-6.07 vs 5.85-5.86 TB/s on two boxes
+// (tools/probe_sgd2.hip round-2 sweep). The pure 2-stream copy keeps
+// ELEM_GRID (6.15 at 1024).
+#define UPD_GRID 768
 // 4-stream fused Add+Get kernels prefer FAT blocks: 1024 threads x 768+
 // blocks measured 6.14 TB/s vs 5.87 at 256x768 (tools/probe_sgd2.hip
 // round-2 sweep; the 2/3-stream kernels regress at 1024 — they keep
@@ -498,7 +503,7 @@ void mv_launch_copy(float* dst, const float* src, int64_t n, hipStream_t s) {
 
 void mv_launch_add(float* data, const float* delta, int64_t n, hipStream_t s) {
   int64_t n4 = n / 4;
-  if (n4) k_add_f4<<<grid_for_cap(n4, ELEM_GRID), BLOCK, 0, s>>>(
+  if (n4) k_add_f4<<<grid_for_cap(n4, UPD_GRID), BLOCK, 0, s>>>(
       (v4f*)data, (const v4f*)delta, n4);
   int64_t tail = n - n4 * 4;
   if (tail) k_add_tail<<<1, 64, 0, s>>>(data, delta, n4 * 4, n);
@@ -506,7 +511,7 @@ void mv_launch_add(float* data, const float* delta, int64_t n, hipStream_t s) {
 
 void mv_launch_sgd(float* data, const float* delta, int64_t n, hipStream_t s) {
   int64_t n4 = n / 4;
-  if (n4) k_sgd_f4<<<grid_for_cap(n4, ELEM_GRID), BLOCK, 0, s>>>(
+  if (n4) k_sgd_f4<<<grid_for_cap(n4, UPD_GRID), BLOCK, 0, s>>>(
       (v4f*)data, (const v4f*)delta, n4);
   int64_t tail = n - n4 * 4;
   if (tail) k_sgd_tail<<<1, 64, 0, s>>>(data, delta, n4 * 4, n);
