@@ -194,14 +194,18 @@ class PSModel:
         batches = [b.to(self.device) for b in batches]
         pulled = self.table.get()
         local = pulled.clone()
-        total_loss = 0.0
         for b in batches:
             CHECK(isinstance(b, DenseBatch),
                   "cfg.sparse=false requires dense batches "
                   "(\"label value value ...\" data)")
-            grad, loss = self.objective.gradient(b, local)
-            total_loss += loss
-            local -= self.sched.next_lr() * grad
+        if local.is_cuda and self._dense_fused_ok():
+            total_loss = self._dense_minibatches_fused(batches, local)
+        else:
+            total_loss = 0.0
+            for b in batches:
+                grad, loss = self.objective.gradient(b, local)
+                total_loss += loss
+                local -= self.sched.next_lr() * grad
         if self.table.updater_type == "adagrad":
             opt = mv.AddOption(learning_rate=1.0,
                                rho=self.cfg.learning_rate)
@@ -209,6 +213,46 @@ class PSModel:
         else:
             self.table.add(pulled - local).wait()
         return total_loss / max(len(batches), 1)
+
+    def _dense_fused_ok(self) -> bool:
+        """GPU dense minibatch fusion applies to the sigmoid/softmax
+        objectives (K <= 64).  FTRL dense is refused upstream."""
+        from .objective import SigmoidObjective, SoftmaxObjective
+        return (type(self.objective) in (SigmoidObjective,
+                                         SoftmaxObjective)
+                and self.cols <= 64)
+
+    def _dense_minibatches_fused(self, batches, local) -> float:
+        """Dense minibatch loop as 4 GPU ops each (GEMM, fused
+        post-kernel, GEMM, update) instead of ~15: scores = X@W stays on
+        rocBLAS/MFMA, then ``lr_dense_post`` turns the logits buffer
+        into the (p - onehot(y))*wt diff IN PLACE and atomically
+        accumulates the per-batch mean loss (one host sync per chunk),
+        then grad = X^T@diff (rocBLAS again) and the SGD/regularizer
+        update runs as in-place torch ops.  Numerics match
+        objective.gradient + the learning-rate step
+        (tests/test_gpu_kernels.py::test_lr_dense_post_*)."""
+        from multiverso_amd import ops as _ops
+        from .objective import L1Regular, L2Regular
+        hip = _ops.module(required=True)
+        reg = self.objective.regular
+        loss_acc = torch.zeros((), device=self.device)
+        for b in batches:
+            B = b.size
+            if B == 0:
+                self.sched.next_lr()
+                continue
+            logits = b.x @ local
+            wts = None if b.weights is None else b.weights.float()
+            hip.lr_dense_post(logits, b.labels.float(), wts, loss_acc,
+                              1.0 / B)
+            lr = self.sched.next_lr()
+            if isinstance(reg, L2Regular):
+                local.mul_(1.0 - lr * reg.coef)
+            elif isinstance(reg, L1Regular):
+                local.add_(torch.sign(local), alpha=-lr * reg.coef)
+            local.addmm_(b.x.t(), logits, alpha=-lr)
+        return float(loss_acc)
 
     def _fused_kind(self, local: torch.Tensor) -> str:
         """Which fused K13/K14 minibatch kernel pair serves this

@@ -57,6 +57,8 @@ void mv_launch_lr_softmax_fwd(const float*, const int64_t*, const float*,
 void mv_launch_lr_softmax_scatter(float*, const int64_t*, const float*,
                                   const int*, const float*, float, int,
                                   float, int64_t, int64_t, hipStream_t);
+void mv_launch_lr_dense_post(float*, const float*, const float*, float*,
+                             float, int64_t, int64_t, hipStream_t);
 void mv_launch_lr_ftrl_fwd(const float*, const int64_t*, const float*,
                            const int*, const float*, const float*, float*,
                            float*, float, float, float, float, int64_t,
@@ -512,6 +514,30 @@ void lr_ftrl_forward(torch::Tensor zn, torch::Tensor keys,
                         cur_stream());
 }
 
+void lr_dense_post(torch::Tensor logits, torch::Tensor labels,
+                   c10::optional<torch::Tensor> wts, torch::Tensor loss_acc,
+                   double inv_b) {
+  check_f32(logits, "logits"); check_f32(labels, "labels");
+  check_f32(loss_acc, "loss_acc");
+  int64_t B = labels.numel();
+  TORCH_CHECK(B > 0 && logits.numel() % B == 0,
+              "logits/labels batch mismatch");
+  int64_t K = logits.numel() / B;
+  TORCH_CHECK(K >= 1 && K <= 64,
+              "fused dense post supports 1..64 classes (got ", K, ")");
+  TORCH_CHECK(loss_acc.numel() == 1, "loss_acc must be a scalar");
+  const float* wp = nullptr;
+  if (wts.has_value()) {
+    check_f32(*wts, "wts");
+    TORCH_CHECK(wts->numel() == B, "weights size mismatch");
+    wp = wts->data_ptr<float>();
+  }
+  mv_launch_lr_dense_post(logits.data_ptr<float>(),
+                          labels.data_ptr<float>(), wp,
+                          loss_acc.data_ptr<float>(), (float)inv_b, B, K,
+                          cur_stream());
+}
+
 void lr_ftrl_scatter(torch::Tensor zn, torch::Tensor keys,
                      torch::Tensor vals, torch::Tensor ptr,
                      torch::Tensor err, double alpha, double beta, double l1,
@@ -566,6 +592,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "err/loss (objective.cpp:193-230)");
   m.def("lr_softmax_scatter", &lr_softmax_scatter,
         "K14 softmax fused: w[key*K+k] -= lr*(val*err_k + reg), atomic");
+  m.def("lr_dense_post", &lr_dense_post,
+        "Dense-mode fused post-GEMM: logits -> softmax/sigmoid diff in "
+        "place + atomic mean-loss accumulate (objective.cpp:193-230 "
+        "dense branch)");
   m.def("lr_ftrl_forward", &lr_ftrl_forward,
         "FTRL fused forward: reconstruct w from (z|n), sigmoid, err/loss "
         "(objective.cpp:250-345)");

@@ -1468,3 +1468,66 @@ void mv_launch_add_i64(int64_t* d, const int64_t* x, int64_t n, hipStream_t s) {
   if (n > 0) k_add_g<int64_t><<<grid_for_cap(n, ELEM_GRID), BLOCK, 0, s>>>(d, x, n);
 }
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// Dense-mode minibatch post-GEMM fusion (reference sparse=false — its
+// mnist.config deployment, objective.cpp:193-230 dense branch).  The two
+// GEMMs (scores = X@W, grad = X^T@diff) stay on rocBLAS/MFMA; this kernel
+// replaces the ~10 elementwise torch ops BETWEEN them (softmax/sigmoid,
+// one_hot, diff, weighting, per-sample loss, loss mean-accumulate) with
+// ONE launch that turns the logits buffer into the diff buffer in place
+// and atomically accumulates mean loss into a chunk-wide scalar (one
+// host sync per chunk).
+//
+// K >= 2: softmax, one wave per sample, lane k owns class k (K <= 64);
+// loss = -log(p_y + eps) (SoftmaxObjective.loss).  K == 1: sigmoid;
+// loss = -(y log p + (1-y) log(1-p)) (base Objective.loss at O=1).
+// ---------------------------------------------------------------------------
+
+__global__ void k_lr_dense_post(float* __restrict__ logits,
+                                const float* __restrict__ labels,
+                                const float* __restrict__ wts,
+                                float* __restrict__ loss_acc,
+                                float inv_b, int B, int K) {
+  int wid = (int)((blockIdx.x * (int64_t)blockDim.x + threadIdx.x) >> 6);
+  int lane = threadIdx.x & 63;
+  int nwaves = (int)((gridDim.x * (int64_t)blockDim.x) >> 6);
+  const float eps = 1e-12f;
+  for (int i = wid; i < B; i += nwaves) {
+    float wt = wts ? wts[i] : 1.f;
+    if (K == 1) {
+      if (lane == 0) {
+        float s = logits[i];
+        float p = 1.f / (1.f + expf(-s));
+        float y = labels[i];
+        logits[i] = (p - y) * wt;
+        float ll = -(y * logf(p + eps) + (1.f - y) * logf(1.f - p + eps));
+        atomicAdd(loss_acc, ll * inv_b);
+      }
+      continue;
+    }
+    float l = (lane < K) ? logits[(int64_t)i * K + lane] : -1e30f;
+    float mx = l;
+#pragma unroll
+    for (int sh = 32; sh; sh >>= 1) mx = fmaxf(mx, __shfl_xor(mx, sh, 64));
+    float e = (lane < K) ? expf(l - mx) : 0.f;
+    float se = e;
+#pragma unroll
+    for (int sh = 32; sh; sh >>= 1) se += __shfl_xor(se, sh, 64);
+    float p = e / se;
+    int y = (int)labels[i];
+    if (lane < K) logits[(int64_t)i * K + lane] = (p - (lane == y)) * wt;
+    float py = __shfl(p, y, 64);
+    if (lane == 0) atomicAdd(loss_acc, -logf(py + eps) * inv_b);
+  }
+}
+
+extern "C" void mv_launch_lr_dense_post(float* logits, const float* labels,
+                                        const float* wts, float* loss_acc,
+                                        float inv_b, int64_t B, int64_t K,
+                                        hipStream_t s) {
+  if (!B) return;
+  k_lr_dense_post<<<grid_for(B * 64), BLOCK, 0, s>>>(logits, labels, wts,
+                                                     loss_acc, inv_b,
+                                                     (int)B, (int)K);
+}
