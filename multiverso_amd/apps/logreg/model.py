@@ -242,10 +242,14 @@ class PSModel:
             if B == 0:
                 self.sched.next_lr()
                 continue
-            logits = b.x @ local
             wts = None if b.weights is None else b.weights.float()
-            hip.lr_dense_post(logits, b.labels.float(), wts, loss_acc,
-                              1.0 / B)
+            logits = torch.empty(B, self.cols, device=self.device)
+            if not hip.lr_dense_fwd(b.x, local, b.labels.float(), wts,
+                                    logits, loss_acc, 1.0 / B):
+                # d*K over the LDS budget: rocBLAS GEMM + fused post
+                torch.matmul(b.x, local, out=logits)
+                hip.lr_dense_post(logits, b.labels.float(), wts, loss_acc,
+                                  1.0 / B)
             lr = self.sched.next_lr()
             if isinstance(reg, L2Regular):
                 local.mul_(1.0 - lr * reg.coef)

@@ -50,6 +50,15 @@ def build(verbose: bool = False) -> str:
             print("+", " ".join(cmd), flush=True)
         subprocess.run(cmd, check=True)
 
+    # cpp_extension's ninja rules only see bindings.cpp; kernels.hip.o is
+    # an opaque ldflag, so a kernel-only change would leave a STALE .so
+    # (observed: an optimized kernel measuring identical to the old one).
+    # Removing the out-of-date .so forces the relink.
+    so = _so_path()
+    if (os.path.exists(so)
+            and os.path.getmtime(kernels_obj) > os.path.getmtime(so)):
+        os.remove(so)
+
     mod = cpp_extension.load(
         name=_MOD_NAME,
         sources=[os.path.join(_CSRC, "bindings.cpp")],

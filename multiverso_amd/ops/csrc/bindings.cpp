@@ -59,6 +59,9 @@ void mv_launch_lr_softmax_scatter(float*, const int64_t*, const float*,
                                   float, int64_t, int64_t, hipStream_t);
 void mv_launch_lr_dense_post(float*, const float*, const float*, float*,
                              float, int64_t, int64_t, hipStream_t);
+int mv_launch_lr_dense_fwd(const float*, const float*, const float*,
+                           const float*, float*, float*, float, int64_t,
+                           int64_t, int64_t, hipStream_t);
 void mv_launch_lr_ftrl_fwd(const float*, const int64_t*, const float*,
                            const int*, const float*, const float*, float*,
                            float*, float, float, float, float, int64_t,
@@ -538,6 +541,32 @@ void lr_dense_post(torch::Tensor logits, torch::Tensor labels,
                           cur_stream());
 }
 
+bool lr_dense_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor labels,
+                  c10::optional<torch::Tensor> wts, torch::Tensor diff,
+                  torch::Tensor loss_acc, double inv_b) {
+  check_f32(x, "x"); check_f32(w, "w"); check_f32(labels, "labels");
+  check_f32(diff, "diff"); check_f32(loss_acc, "loss_acc");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(0),
+              "x [B,d] and w [d,K] shape mismatch");
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous(), "x/w must be "
+              "contiguous");
+  int64_t B = x.size(0), d = x.size(1), K = w.size(1);
+  TORCH_CHECK(labels.numel() == B, "labels batch mismatch");
+  TORCH_CHECK(diff.numel() == B * K, "diff must be [B, K]");
+  TORCH_CHECK(loss_acc.numel() == 1, "loss_acc must be a scalar");
+  const float* wp = nullptr;
+  if (wts.has_value()) {
+    check_f32(*wts, "wts");
+    TORCH_CHECK(wts->numel() == B, "weights size mismatch");
+    wp = wts->data_ptr<float>();
+  }
+  return mv_launch_lr_dense_fwd(
+             x.data_ptr<float>(), w.data_ptr<float>(),
+             labels.data_ptr<float>(), wp, diff.data_ptr<float>(),
+             loss_acc.data_ptr<float>(), (float)inv_b, B, d, K,
+             cur_stream()) != 0;
+}
+
 void lr_ftrl_scatter(torch::Tensor zn, torch::Tensor keys,
                      torch::Tensor vals, torch::Tensor ptr,
                      torch::Tensor err, double alpha, double beta, double l1,
@@ -592,6 +621,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "err/loss (objective.cpp:193-230)");
   m.def("lr_softmax_scatter", &lr_softmax_scatter,
         "K14 softmax fused: w[key*K+k] -= lr*(val*err_k + reg), atomic");
+  m.def("lr_dense_fwd", &lr_dense_fwd,
+        "Dense-mode fused forward: X@W + softmax/sigmoid diff + loss in "
+        "one kernel, W LDS-staged; returns False when d*K exceeds the "
+        "LDS budget (caller uses the GEMM + lr_dense_post path)");
   m.def("lr_dense_post", &lr_dense_post,
         "Dense-mode fused post-GEMM: logits -> softmax/sigmoid diff in "
         "place + atomic mean-loss accumulate (objective.cpp:193-230 "

@@ -1489,10 +1489,18 @@ __global__ void k_lr_dense_post(float* __restrict__ logits,
                                 const float* __restrict__ wts,
                                 float* __restrict__ loss_acc,
                                 float inv_b, int B, int K) {
+  // Loss goes lane-private -> one LDS atomic per wave -> ONE global
+  // atomic per block: a naive per-sample atomicAdd on the single
+  // loss_acc cache line serialized 4096 waves and cost 54 us (measured,
+  // profiles/round2_dense_fused.md) — 13x the kernel's real work.
+  __shared__ float block_loss;
+  if (threadIdx.x == 0) block_loss = 0.f;
+  __syncthreads();
   int wid = (int)((blockIdx.x * (int64_t)blockDim.x + threadIdx.x) >> 6);
   int lane = threadIdx.x & 63;
   int nwaves = (int)((gridDim.x * (int64_t)blockDim.x) >> 6);
   const float eps = 1e-12f;
+  float ll_sum = 0.f;
   for (int i = wid; i < B; i += nwaves) {
     float wt = wts ? wts[i] : 1.f;
     if (K == 1) {
@@ -1501,8 +1509,7 @@ __global__ void k_lr_dense_post(float* __restrict__ logits,
         float p = 1.f / (1.f + expf(-s));
         float y = labels[i];
         logits[i] = (p - y) * wt;
-        float ll = -(y * logf(p + eps) + (1.f - y) * logf(1.f - p + eps));
-        atomicAdd(loss_acc, ll * inv_b);
+        ll_sum -= y * logf(p + eps) + (1.f - y) * logf(1.f - p + eps);
       }
       continue;
     }
@@ -1518,8 +1525,12 @@ __global__ void k_lr_dense_post(float* __restrict__ logits,
     int y = (int)labels[i];
     if (lane < K) logits[(int64_t)i * K + lane] = (p - (lane == y)) * wt;
     float py = __shfl(p, y, 64);
-    if (lane == 0) atomicAdd(loss_acc, -logf(py + eps) * inv_b);
+    if (lane == 0) ll_sum -= logf(py + eps);
   }
+  if (lane == 0 && ll_sum != 0.f) atomicAdd(&block_loss, ll_sum);
+  __syncthreads();
+  if (threadIdx.x == 0 && block_loss != 0.f)
+    atomicAdd(loss_acc, block_loss * inv_b);
 }
 
 extern "C" void mv_launch_lr_dense_post(float* logits, const float* labels,
@@ -1527,7 +1538,128 @@ extern "C" void mv_launch_lr_dense_post(float* logits, const float* labels,
                                         float inv_b, int64_t B, int64_t K,
                                         hipStream_t s) {
   if (!B) return;
-  k_lr_dense_post<<<grid_for(B * 64), BLOCK, 0, s>>>(logits, labels, wts,
-                                                     loss_acc, inv_b,
-                                                     (int)B, (int)K);
+  k_lr_dense_post<<<grid_for_cap(B * 64, 512), BLOCK, 0, s>>>(
+      logits, labels, wts, loss_acc, inv_b, (int)B, (int)K);
+}
+
+// ---------------------------------------------------------------------------
+// Dense-mode fused FORWARD (scores = X@W -> softmax/sigmoid -> diff +
+// loss) in ONE kernel: W staged to LDS once per workgroup (d*K*4 B,
+// odd-padded row stride so the per-element class reads are LDS
+// bank-conflict-free), one 64-lane wave per sample, lane e-strided over
+// the X row (coalesced 256 B/instr stream — X is read EXACTLY ONCE, the
+// memory floor for this op).  Replaces the rocBLAS skinny GEMM
+// [B,d]x[d,K<=16] whose 16x16 macro-tiles ran at 1.3 TB/s.
+//
+// K is an EXACT compile-time parameter (one instantiation per class
+// count 1..16).  The first version carried a runtime `if (k < K)` guard
+// inside the unrolled class loop; the compiler turned that into a
+// per-class branch ladder with a full `s_waitcnt vmcnt(0) lgkmcnt(0)`
+// after EVERY ds_read_b32 — 102 us measured (5x SLOWER than the rocBLAS
+// pair it meant to replace).  Branch-free exact-K bodies batch the K
+// LDS reads under one waitcnt.  Falls back to rocBLAS +
+// k_lr_dense_post when K > 16 or d*(K|1)*4 exceeds the LDS budget.
+// ---------------------------------------------------------------------------
+
+template <int K>
+__global__ void k_lr_dense_fwd(const float* __restrict__ X,
+                               const float* __restrict__ W,
+                               const float* __restrict__ labels,
+                               const float* __restrict__ wts,
+                               float* __restrict__ diff,
+                               float* __restrict__ loss_acc,
+                               float inv_b, int B, int d) {
+  extern __shared__ float wlds[];  // [d][K|1]
+  __shared__ float block_loss;
+  constexpr int S = K | 1;  // odd stride: (S*lane) % 64 banks all distinct
+  for (int t = threadIdx.x; t < d * K; t += blockDim.x)
+    wlds[(t / K) * S + (t % K)] = W[t];
+  if (threadIdx.x == 0) block_loss = 0.f;
+  __syncthreads();
+
+  int wid = (int)((blockIdx.x * (int64_t)blockDim.x + threadIdx.x) >> 6);
+  int lane = threadIdx.x & 63;
+  int nwaves = (int)((gridDim.x * (int64_t)blockDim.x) >> 6);
+  const float eps = 1e-12f;
+  float ll_sum = 0.f;
+  for (int i = wid; i < B; i += nwaves) {
+    const float* __restrict__ xrow = X + (int64_t)i * d;
+    float acc[K];
+#pragma unroll
+    for (int k = 0; k < K; ++k) acc[k] = 0.f;
+    for (int e = lane; e < d; e += 64) {
+      float x = xrow[e];
+      const float* wr = wlds + e * S;
+#pragma unroll
+      for (int k = 0; k < K; ++k) acc[k] += x * wr[k];
+    }
+#pragma unroll
+    for (int k = 0; k < K; ++k)
+#pragma unroll
+      for (int sh = 32; sh; sh >>= 1) acc[k] += __shfl_xor(acc[k], sh, 64);
+    float wt = wts ? wts[i] : 1.f;
+    if (K == 1) {
+      if (lane == 0) {
+        float p = 1.f / (1.f + expf(-acc[0]));
+        float y = labels[i];
+        diff[i] = (p - y) * wt;
+        ll_sum -= y * logf(p + eps) + (1.f - y) * logf(1.f - p + eps);
+      }
+      continue;
+    }
+    // lane k owns class k (logits identical on all lanes after the
+    // reductions); unrolled select keeps acc[] in registers
+    float mine = -1e30f;
+#pragma unroll
+    for (int k = 0; k < K; ++k)
+      if (lane == k) mine = acc[k];
+    float mx = mine;
+#pragma unroll
+    for (int sh = 32; sh; sh >>= 1) mx = fmaxf(mx, __shfl_xor(mx, sh, 64));
+    float e = (lane < K) ? expf(mine - mx) : 0.f;
+    float se = e;
+#pragma unroll
+    for (int sh = 32; sh; sh >>= 1) se += __shfl_xor(se, sh, 64);
+    float p = e / se;
+    int y = (int)labels[i];
+    if (lane < K) diff[(int64_t)i * K + lane] = (p - (lane == y)) * wt;
+    float py = __shfl(p, y, 64);
+    if (lane == 0) ll_sum -= logf(py + eps);
+  }
+  if (lane == 0 && ll_sum != 0.f) atomicAdd(&block_loss, ll_sum);
+  __syncthreads();
+  if (threadIdx.x == 0 && block_loss != 0.f)
+    atomicAdd(loss_acc, block_loss * inv_b);
+}
+
+extern "C" int mv_launch_lr_dense_fwd(const float* X, const float* W,
+                                      const float* labels, const float* wts,
+                                      float* diff, float* loss_acc,
+                                      float inv_b, int64_t B, int64_t d,
+                                      int64_t K, hipStream_t s) {
+  if (!B) return 1;
+  size_t lds = (size_t)d * (K | 1) * sizeof(float);
+  if (K > 16 || lds > 144 * 1024) return 0;  // rocBLAS + post fallback
+  // The W staging leaves LDS room for only ~1 WG/CU, so the workgroup
+  // carries the CU's whole latency-hiding budget itself: 1024 threads
+  // = 16 waves, one workgroup per CU, one staging pass per CU.
+  int64_t blocks = (B + 15) / 16;  // 16 waves/block, one sample per wave
+  if (blocks > 256) blocks = 256;  // 1 LDS-heavy WG per CU
+  int grid = (int)blocks;
+  switch (K) {
+#define LAUNCH_KT(KT)                                                      \
+  case KT:                                                                 \
+    k_lr_dense_fwd<KT><<<grid, COPY_BLOCK, lds, s>>>(X, W, labels, wts,    \
+                                                     diff, loss_acc,       \
+                                                     inv_b, (int)B,        \
+                                                     (int)d);              \
+    break
+    LAUNCH_KT(1); LAUNCH_KT(2); LAUNCH_KT(3); LAUNCH_KT(4);
+    LAUNCH_KT(5); LAUNCH_KT(6); LAUNCH_KT(7); LAUNCH_KT(8);
+    LAUNCH_KT(9); LAUNCH_KT(10); LAUNCH_KT(11); LAUNCH_KT(12);
+    LAUNCH_KT(13); LAUNCH_KT(14); LAUNCH_KT(15); LAUNCH_KT(16);
+#undef LAUNCH_KT
+    default: return 0;
+  }
+  return 1;
 }

@@ -774,3 +774,55 @@ def test_dense_fused_chunk_vs_torch(objective, K, reg):
         (l_fused - l_ref).abs().max()
     assert abs(loss_fused - loss_ref) < 1e-3 * max(1.0, abs(loss_ref))
     mv.shutdown()
+
+
+@pytest.mark.parametrize("K,d,weighted", [(1, 512, True), (10, 2048, False),
+                                          (10, 2048, True), (16, 1000, False)])
+def test_lr_dense_fwd_vs_torch(hip, K, d, weighted):
+    """Fused dense forward (X@W + softmax/sigmoid + diff + loss, W in
+    LDS) vs the two-step torch reference."""
+    torch.manual_seed(23 + K)
+    B = 301
+    x = torch.randn(B, d, device="cuda:0")
+    w = (torch.randn(d, K, device="cuda:0") * 0.05)
+    wts = (torch.rand(B, device="cuda:0") + 0.5) if weighted else None
+    eps = 1e-12
+    logits = x @ w
+    if K == 1:
+        labels = torch.randint(0, 2, (B,)).float().cuda()
+        p = torch.sigmoid(logits)
+        y = labels.unsqueeze(1)
+        ref_diff = p - y
+        ref_loss = float(-(y * torch.log(p + eps)
+                           + (1 - y) * torch.log(1 - p + eps)).sum(1).mean())
+    else:
+        labels = torch.randint(0, K, (B,)).float().cuda()
+        p = torch.softmax(logits, dim=1)
+        onehot = torch.nn.functional.one_hot(labels.long(), K).float()
+        ref_diff = p - onehot
+        ref_loss = float(-torch.log(
+            p[torch.arange(B, device="cuda:0"), labels.long()]
+            + eps).mean())
+    if wts is not None:
+        ref_diff = ref_diff * wts.unsqueeze(1)
+
+    diff = torch.empty(B, K, device="cuda:0")
+    loss_acc = torch.zeros((), device="cuda:0")
+    took = hip.lr_dense_fwd(x, w, labels, wts, diff, loss_acc, 1.0 / B)
+    torch.cuda.synchronize()
+    assert took, "fused fwd refused a d*K that fits LDS"
+    assert torch.allclose(diff, ref_diff, rtol=1e-4, atol=1e-5), \
+        (diff - ref_diff).abs().max()
+    assert abs(float(loss_acc) - ref_loss) < 1e-3 * max(1.0, abs(ref_loss))
+
+
+def test_lr_dense_fwd_lds_fallback(hip):
+    """d*K beyond the LDS budget must return False (caller then uses the
+    rocBLAS + lr_dense_post path), not launch a broken kernel."""
+    B, d, K = 8, 2048, 64  # 2048*65*4 = 520 KB >> 144 KB budget
+    x = torch.randn(B, d, device="cuda:0")
+    w = torch.randn(d, K, device="cuda:0")
+    labels = torch.randint(0, K, (B,)).float().cuda()
+    diff = torch.empty(B, K, device="cuda:0")
+    loss_acc = torch.zeros((), device="cuda:0")
+    assert not hip.lr_dense_fwd(x, w, labels, None, diff, loss_acc, 1.0 / B)
