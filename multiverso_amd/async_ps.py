@@ -112,13 +112,28 @@ class AsyncEngine:
     # ------------------------------------------------------------------
     # worker side
     # ------------------------------------------------------------------
+    # Request wire format: a fixed-size header message + AT MOST ONE
+    # payload blob message per request (each gloo message costs ~0.2 ms
+    # of latency on the host lane — round-2 measurement in
+    # docs/ENGINEERING_NOTES.md — so an Add is 2 messages, not 3-4).
+    # Blob layout, 8-byte aligned by construction:
+    #   [opt: 5 f64 if has_opt][keys: n_keys i64][vals: n_vals dtype]
     def _send_request(self, dst: int, hdr_fields: List[int],
-                      payloads: List[torch.Tensor]) -> List:
+                      payloads: List[torch.Tensor]):
+        """Returns (works, refs): pending isends + the buffers that must
+        outlive them."""
         hdr = torch.tensor(hdr_fields, dtype=torch.int64)
         works = [dist.isend(hdr, dst, group=self.req)]
-        for p in payloads:
-            works.append(dist.isend(p, dst, group=self.req))
-        return works
+        refs: List[torch.Tensor] = [hdr]
+        if payloads:
+            if len(payloads) == 1:
+                blob = payloads[0].contiguous().view(torch.uint8).view(-1)
+            else:
+                blob = torch.cat([p.contiguous().view(torch.uint8).view(-1)
+                                  for p in payloads])
+            works.append(dist.isend(blob, dst, group=self.req))
+            refs.append(blob)
+        return works, refs
 
     def whole_add(self, table, delta_flat: torch.Tensor, unit: int,
                   option: Optional[AddOption], want_ack: bool):
@@ -139,9 +154,10 @@ class AsyncEngine:
             payload = piece.cpu().contiguous()
             hdr = [OP_ADD, table.table_id, 0, payload.numel(),
                    1 if want_ack else 0, 0 if opt is None else 1]
-            ps = [payload] + ([] if opt is None else [opt])
-            works += self._send_request(dst, hdr, ps)
-            refs += ps
+            ps = ([] if opt is None else [opt]) + [payload]
+            w, r = self._send_request(dst, hdr, ps)
+            works += w
+            refs += r
             if want_ack:
                 ack = torch.empty(1, dtype=torch.int64)
                 works.append(dist.irecv(ack, dst, group=self.rep))
@@ -163,7 +179,9 @@ class AsyncEngine:
                 table._server_read_chunk_into(dst_slice)
                 continue
             hdr = [OP_GET, table.table_id, 0, cnt * unit, 0, 0]
-            works += self._send_request(dst, hdr, [])
+            w, r = self._send_request(dst, hdr, [])
+            works += w
+            refs += r
             buf = torch.empty(cnt * unit, dtype=table.dtype)
             works.append(dist.irecv(buf, dst, group=self.rep))
             slots.append((buf, dst_slice))
@@ -208,9 +226,10 @@ class AsyncEngine:
             kcpu = keys.contiguous()
             hdr = [OP_ADD_ROWS, table.table_id, cnt, payload.numel(), 0,
                    0 if opt is None else 1]
-            ps = [kcpu, payload] + ([] if opt is None else [opt])
-            works += self._send_request(dst, hdr, ps)
-            refs += ps
+            ps = ([] if opt is None else [opt]) + [kcpu, payload]
+            w, r = self._send_request(dst, hdr, ps)
+            works += w
+            refs += r
         return PendingReply(works, refs)
 
     def keyed_get(self, table, ids_cpu: torch.Tensor,
@@ -241,11 +260,13 @@ class AsyncEngine:
                 continue
             kcpu = keys.contiguous()
             hdr = [OP_GET_ROWS, table.table_id, cnt, cnt * unit, 0, 0]
-            works += self._send_request(dst, hdr, [kcpu])
+            w, r = self._send_request(dst, hdr, [kcpu])
+            works += w
+            refs += r
             buf = torch.empty(cnt * unit, dtype=table.dtype)
             works.append(dist.irecv(buf, dst, group=self.rep))
             slots.append((buf, rows))
-            refs += [kcpu, buf]
+            refs.append(buf)
         for w in works:
             w.wait()
         for buf, rows in slots:
@@ -268,8 +289,9 @@ class AsyncEngine:
                 table._server_kv_add(k, v)
                 continue
             hdr = [OP_KV_ADD, table.table_id, k.numel(), v.numel(), 0, 0]
-            works += self._send_request(dst, hdr, [k, v])
-            refs += [k, v]
+            w, r = self._send_request(dst, hdr, [k, v])
+            works += w
+            refs += r
         return PendingReply(works, refs)
 
     def kv_get(self, table, keys: torch.Tensor) -> torch.Tensor:
@@ -277,6 +299,7 @@ class AsyncEngine:
         n = table.num_shards
         out = torch.zeros(keys.numel(), dtype=torch.float64)
         works: List = []
+        refs: List = []
         slots = []
         for s in range(n):
             m = (keys % n) == s
@@ -288,7 +311,9 @@ class AsyncEngine:
                 out[m] = table._server_kv_get(k)
                 continue
             hdr = [OP_KV_GET, table.table_id, k.numel(), k.numel(), 0, 0]
-            works += self._send_request(dst, hdr, [k])
+            w, r = self._send_request(dst, hdr, [k])
+            works += w
+            refs += r
             buf = torch.empty(k.numel(), dtype=torch.float64)
             works.append(dist.irecv(buf, dst, group=self.rep))
             slots.append((buf, m))
@@ -319,8 +344,9 @@ class AsyncEngine:
                     total += ids.numel()
                 continue
             hdr = [OP_GET_STALE, table.table_id, 0, 0, 0, 0]
-            for w in self._send_request(dst, hdr, []):
-                w.wait()
+            w, _r = self._send_request(dst, hdr, [])
+            for ww in w:
+                ww.wait()
             remotes.append(dst)
         for dst in remotes:
             cnt = torch.empty(1, dtype=torch.int64)
@@ -428,22 +454,32 @@ class AsyncEngine:
     def _serve_one(self, src: int, op: int, tid: int, n_keys: int,
                    n_vals: int, want_ack: bool, has_opt: bool) -> None:
         table = self._table(tid)
+        # One blob recv mirrors the worker's one blob send (layout in
+        # _send_request: [opt][keys][vals], 8-byte aligned sections)
+        vals_dtype = None
+        if n_vals and op in (OP_ADD, OP_ADD_ROWS):
+            vals_dtype = table.dtype
+        elif n_vals and op == OP_KV_ADD:
+            vals_dtype = torch.float64
+        vals_bytes = (0 if vals_dtype is None
+                      else n_vals * torch.empty(0, dtype=vals_dtype)
+                                         .element_size())
+        nbytes = (40 if has_opt else 0) + n_keys * 8 + vals_bytes
         keys = None
         vals = None
-        if n_keys:
-            keys = torch.empty(n_keys, dtype=torch.int64)
-            dist.recv(keys, src, group=self.req)
-        if n_vals and op in (OP_ADD, OP_ADD_ROWS):
-            vals = torch.empty(n_vals, dtype=table.dtype)
-            dist.recv(vals, src, group=self.req)
-        elif n_vals and op == OP_KV_ADD:
-            vals = torch.empty(n_vals, dtype=torch.float64)
-            dist.recv(vals, src, group=self.req)
         option = None
-        if has_opt:
-            ot = torch.empty(5, dtype=torch.float64)
-            dist.recv(ot, src, group=self.req)
-            option = _opt_from(ot)
+        if nbytes:
+            blob = torch.empty(nbytes, dtype=torch.uint8)
+            dist.recv(blob, src, group=self.req)
+            off = 0
+            if has_opt:
+                option = _opt_from(blob[:40].view(torch.float64))
+                off = 40
+            if n_keys:
+                keys = blob[off:off + n_keys * 8].view(torch.int64)
+                off += n_keys * 8
+            if vals_dtype is not None:
+                vals = blob[off:off + vals_bytes].view(vals_dtype)
 
         if op == OP_ADD:
             with monitor("server.process_add"):
