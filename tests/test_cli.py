@@ -195,3 +195,39 @@ def test_selftest_dispatcher_ws2(name, tmp_path):
         capture_output=True, text=True, timeout=300, env=env)
     assert r.returncode == 0, (name, r.stdout[-500:], r.stderr[-1500:])
     assert "PASS" in r.stdout
+
+
+@pytest.mark.parametrize("appargs", [
+    ["--rows", "4000", "--cols", "8", "--steps", "3", "--warmup", "1",
+     "--phase-steps", "2"],
+    ["--app", "logreg", "--steps", "2", "--warmup", "1"],
+    ["--app", "wordembedding", "--steps", "2", "--warmup", "1",
+     "--block-words", "2000", "--vocab", "200", "--dim", "16"],
+    ["--app", "sweep", "--gb", "0.001", "--steps", "2", "--warmup", "1"],
+])
+def test_bench_rehearsal_ws2(appargs, tmp_path):
+    """Rehearse the EXACT driver invocation shape (torchrun --nnodes=1
+    --nproc-per-node N bench.py ...) at world_size 2 on gloo/CPU: every
+    N>1-only branch of every bench app must run and rank 0 must print a
+    parseable JSON contract line.  The driver's SCALE run gets no second
+    chance — this is its CI net."""
+    import json as _json
+    import subprocess, sys, os
+    from conftest import free_port
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + ":" + env.get("PYTHONPATH", "")
+    env["HIP_VISIBLE_DEVICES"] = env["CUDA_VISIBLE_DEVICES"] = ""
+    env["MV_BACKEND"] = "gloo"
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(free_port()),
+         os.path.join(REPO, "bench.py"), "--gpus", "2"] + appargs,
+        capture_output=True, text=True, timeout=600, env=env)
+    assert r.returncode == 0, (r.stdout[-800:], r.stderr[-1500:])
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert line, r.stdout[-800:]
+    rec = _json.loads(line[-1])
+    assert rec["n_gpus"] == 2 and "value" in rec and "metric" in rec
+    if "--app" not in appargs:
+        assert "phases" in rec and rec["phases"], rec
