@@ -61,7 +61,7 @@ void mv_launch_lr_dense_post(float*, const float*, const float*, float*,
                              float, int64_t, int64_t, hipStream_t);
 int mv_launch_lr_dense_fwd(const float*, const float*, const float*,
                            const float*, float*, float*, float, int64_t,
-                           int64_t, int64_t, hipStream_t);
+                           int64_t, int64_t, int, hipStream_t);
 void mv_launch_lr_ftrl_fwd(const float*, const int64_t*, const float*,
                            const int*, const float*, const float*, float*,
                            float*, float, float, float, float, int64_t,
@@ -543,7 +543,7 @@ void lr_dense_post(torch::Tensor logits, torch::Tensor labels,
 
 bool lr_dense_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor labels,
                   c10::optional<torch::Tensor> wts, torch::Tensor diff,
-                  torch::Tensor loss_acc, double inv_b) {
+                  torch::Tensor loss_acc, double inv_b, bool nt) {
   check_f32(x, "x"); check_f32(w, "w"); check_f32(labels, "labels");
   check_f32(diff, "diff"); check_f32(loss_acc, "loss_acc");
   TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(0),
@@ -564,7 +564,7 @@ bool lr_dense_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor labels,
              x.data_ptr<float>(), w.data_ptr<float>(),
              labels.data_ptr<float>(), wp, diff.data_ptr<float>(),
              loss_acc.data_ptr<float>(), (float)inv_b, B, d, K,
-             cur_stream()) != 0;
+             nt ? 1 : 0, cur_stream()) != 0;
 }
 
 void lr_ftrl_scatter(torch::Tensor zn, torch::Tensor keys,
@@ -624,7 +624,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lr_dense_fwd", &lr_dense_fwd,
         "Dense-mode fused forward: X@W + softmax/sigmoid diff + loss in "
         "one kernel, W LDS-staged; returns False when d*K exceeds the "
-        "LDS budget (caller uses the GEMM + lr_dense_post path)");
+        "LDS budget (caller uses the GEMM + lr_dense_post path)",
+        py::arg("x"), py::arg("w"), py::arg("labels"), py::arg("wts"),
+        py::arg("diff"), py::arg("loss_acc"), py::arg("inv_b"),
+        py::arg("nt") = true);
   m.def("lr_dense_post", &lr_dense_post,
         "Dense-mode fused post-GEMM: logits -> softmax/sigmoid diff in "
         "place + atomic mean-loss accumulate (objective.cpp:193-230 "

@@ -457,6 +457,34 @@ __global__ void k_row_gather(float* __restrict__ out,
   }
 }
 
+// u32 loop-index variant: a 64-bit divide is ~40 cycles per element and
+// these gathers run one divide per 4 B moved.  Only the (i -> r, c)
+// math narrows — the shard offset stays 64-bit (1e9-row tables).
+// Launcher picks this whenever total fits u32 (all real pulls do).
+__global__ void k_row_gather_u32(float* __restrict__ out,
+                                 const float* __restrict__ shard,
+                                 const int64_t* __restrict__ rows,
+                                 uint32_t total, uint32_t cols) {
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    uint32_t r = i / cols, c = i - r * cols;
+    out[i] = shard[rows[r] * (int64_t)cols + c];
+  }
+}
+
+__global__ void k_row_gather_f2_u32(float2* __restrict__ out,
+                                    const float2* __restrict__ shard,
+                                    const int64_t* __restrict__ rows,
+                                    uint32_t total2, uint32_t cols2) {
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total2;
+       i += stride) {
+    uint32_t r = i / cols2, c = i - r * cols2;
+    out[i] = shard[rows[r] * (int64_t)cols2 + c];
+  }
+}
+
 __global__ void k_row_scatter_add(float* __restrict__ shard,
                                   const float* __restrict__ vals,
                                   const int64_t* __restrict__ rows,
@@ -482,6 +510,35 @@ __global__ void k_row_scatter_add_u(float* __restrict__ shard,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
     int64_t r = i / cols, c = i % cols;
     int64_t k = rows[r] * cols + c;
+    shard[k] += sign * vals[i];
+  }
+}
+
+// u32 loop-index forms (see k_row_gather_u32: one 64-bit divide per 4 B
+// is real VALU time on a gather/scatter; the shard offset stays 64-bit)
+__global__ void k_row_scatter_add_u32(float* __restrict__ shard,
+                                      const float* __restrict__ vals,
+                                      const int64_t* __restrict__ rows,
+                                      float sign, uint32_t total,
+                                      uint32_t cols) {
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    uint32_t r = i / cols, c = i - r * cols;
+    atomicAdd(&shard[rows[r] * (int64_t)cols + c], sign * vals[i]);
+  }
+}
+
+__global__ void k_row_scatter_add_u_u32(float* __restrict__ shard,
+                                        const float* __restrict__ vals,
+                                        const int64_t* __restrict__ rows,
+                                        float sign, uint32_t total,
+                                        uint32_t cols) {
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    uint32_t r = i / cols, c = i - r * cols;
+    int64_t k = rows[r] * (int64_t)cols + c;
     shard[k] += sign * vals[i];
   }
 }
@@ -623,12 +680,20 @@ void mv_launch_dcasgda(float* data, float* bak, float* msq, const float* delta,
 void mv_launch_row_gather(float* out, const float* shard, const int64_t* rows,
                           int64_t nrows, int64_t cols, hipStream_t s) {
   if (!nrows || !cols) return;
+  int64_t total = nrows * cols;
   if (cols % 4 == 0) {
     int64_t cols4 = cols / 4;
     k_row_gather_f4<<<grid_for(nrows * cols4), BLOCK, 0, s>>>(
         (float4*)out, (const float4*)shard, rows, nrows, cols4);
+  } else if (cols % 2 == 0 && total / 2 <= UINT32_MAX) {
+    k_row_gather_f2_u32<<<grid_for(total / 2), BLOCK, 0, s>>>(
+        (float2*)out, (const float2*)shard, rows, (uint32_t)(total / 2),
+        (uint32_t)(cols / 2));
+  } else if (total <= UINT32_MAX) {
+    k_row_gather_u32<<<grid_for(total), BLOCK, 0, s>>>(
+        out, shard, rows, (uint32_t)total, (uint32_t)cols);
   } else {
-    k_row_gather<<<grid_for(nrows * cols), BLOCK, 0, s>>>(
+    k_row_gather<<<grid_for(total), BLOCK, 0, s>>>(
         out, shard, rows, nrows, cols);
   }
 }
@@ -638,12 +703,21 @@ void mv_launch_row_scatter_add(float* shard, const float* vals,
                                int64_t nrows, int64_t cols,
                                int assume_unique, hipStream_t s) {
   if (!nrows || !cols) return;
-  if (assume_unique)
-    k_row_scatter_add_u<<<grid_for(nrows * cols), BLOCK, 0, s>>>(
+  int64_t total = nrows * cols;
+  if (total <= UINT32_MAX) {
+    if (assume_unique)
+      k_row_scatter_add_u_u32<<<grid_for(total), BLOCK, 0, s>>>(
+          shard, vals, rows, sign, (uint32_t)total, (uint32_t)cols);
+    else
+      k_row_scatter_add_u32<<<grid_for(total), BLOCK, 0, s>>>(
+          shard, vals, rows, sign, (uint32_t)total, (uint32_t)cols);
+  } else if (assume_unique) {
+    k_row_scatter_add_u<<<grid_for(total), BLOCK, 0, s>>>(
         shard, vals, rows, sign, nrows, cols);
-  else
-    k_row_scatter_add<<<grid_for(nrows * cols), BLOCK, 0, s>>>(
+  } else {
+    k_row_scatter_add<<<grid_for(total), BLOCK, 0, s>>>(
         shard, vals, rows, sign, nrows, cols);
+  }
 }
 
 }  // extern "C"
@@ -1098,17 +1172,70 @@ __global__ void k_row_scatter_adagrad_u(float* __restrict__ shard,
   }
 }
 
+// u32 loop-index forms (one 64-bit divide per element is real VALU
+// time; shard/gsq offsets stay 64-bit for 1e9-row tables)
+__global__ void k_row_scatter_adagrad_u32(float* __restrict__ shard,
+                                          float* __restrict__ gsq,
+                                          const float* __restrict__ vals,
+                                          const int64_t* __restrict__ rows,
+                                          float inv_lr, float rho, float eps,
+                                          uint32_t total, uint32_t cols) {
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    uint32_t r = i / cols, c = i - r * cols;
+    int64_t k = rows[r] * (int64_t)cols + c;
+    float g = vals[i] * inv_lr;
+    if (g != 0.0f) {
+      float G = atomicAdd(&gsq[k], g * g) + g * g;
+      atomicAdd(&shard[k], -rho * g * __frsqrt_rn(G + eps));
+    }
+  }
+}
+
+__global__ void k_row_scatter_adagrad_u_u32(float* __restrict__ shard,
+                                            float* __restrict__ gsq,
+                                            const float* __restrict__ vals,
+                                            const int64_t* __restrict__ rows,
+                                            float inv_lr, float rho,
+                                            float eps, uint32_t total,
+                                            uint32_t cols) {
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    uint32_t r = i / cols, c = i - r * cols;
+    int64_t k = rows[r] * (int64_t)cols + c;
+    float g = vals[i] * inv_lr;
+    if (g != 0.0f) {
+      float G = gsq[k] + g * g;
+      gsq[k] = G;
+      shard[k] -= rho * g * __frsqrt_rn(G + eps);
+    }
+  }
+}
+
 extern "C" void mv_launch_row_scatter_adagrad(
     float* shard, float* gsq, const float* vals, const int64_t* rows,
     float lr, float rho, float eps, int64_t nrows, int64_t cols,
     int assume_unique, hipStream_t s) {
   if (!nrows || !cols) return;
-  if (assume_unique)
-    k_row_scatter_adagrad_u<<<grid_for(nrows * cols), BLOCK, 0, s>>>(
+  int64_t total = nrows * cols;
+  if (total <= UINT32_MAX) {
+    if (assume_unique)
+      k_row_scatter_adagrad_u_u32<<<grid_for(total), BLOCK, 0, s>>>(
+          shard, gsq, vals, rows, 1.0f / lr, rho, eps, (uint32_t)total,
+          (uint32_t)cols);
+    else
+      k_row_scatter_adagrad_u32<<<grid_for(total), BLOCK, 0, s>>>(
+          shard, gsq, vals, rows, 1.0f / lr, rho, eps, (uint32_t)total,
+          (uint32_t)cols);
+  } else if (assume_unique) {
+    k_row_scatter_adagrad_u<<<grid_for(total), BLOCK, 0, s>>>(
         shard, gsq, vals, rows, 1.0f / lr, rho, eps, nrows, cols);
-  else
-    k_row_scatter_adagrad<<<grid_for(nrows * cols), BLOCK, 0, s>>>(
+  } else {
+    k_row_scatter_adagrad<<<grid_for(total), BLOCK, 0, s>>>(
         shard, gsq, vals, rows, 1.0f / lr, rho, eps, nrows, cols);
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -1561,7 +1688,7 @@ extern "C" void mv_launch_lr_dense_post(float* logits, const float* labels,
 // k_lr_dense_post when K > 16 or d*(K|1)*4 exceeds the LDS budget.
 // ---------------------------------------------------------------------------
 
-template <int K>
+template <int K, bool NT>
 __global__ void k_lr_dense_fwd(const float* __restrict__ X,
                                const float* __restrict__ W,
                                const float* __restrict__ labels,
@@ -1588,7 +1715,7 @@ __global__ void k_lr_dense_fwd(const float* __restrict__ X,
 #pragma unroll
     for (int k = 0; k < K; ++k) acc[k] = 0.f;
     for (int e = lane; e < d; e += 64) {
-      float x = xrow[e];
+      float x = NT ? __builtin_nontemporal_load(xrow + e) : xrow[e];
       const float* wr = wlds + e * S;
 #pragma unroll
       for (int k = 0; k < K; ++k) acc[k] += x * wr[k];
@@ -1636,7 +1763,7 @@ extern "C" int mv_launch_lr_dense_fwd(const float* X, const float* W,
                                       const float* labels, const float* wts,
                                       float* diff, float* loss_acc,
                                       float inv_b, int64_t B, int64_t d,
-                                      int64_t K, hipStream_t s) {
+                                      int64_t K, int nt, hipStream_t s) {
   if (!B) return 1;
   size_t lds = (size_t)d * (K | 1) * sizeof(float);
   if (K > 16 || lds > 144 * 1024) return 0;  // rocBLAS + post fallback
@@ -1649,10 +1776,12 @@ extern "C" int mv_launch_lr_dense_fwd(const float* X, const float* W,
   switch (K) {
 #define LAUNCH_KT(KT)                                                      \
   case KT:                                                                 \
-    k_lr_dense_fwd<KT><<<grid, COPY_BLOCK, lds, s>>>(X, W, labels, wts,    \
-                                                     diff, loss_acc,       \
-                                                     inv_b, (int)B,        \
-                                                     (int)d);              \
+    if (nt)                                                                \
+      k_lr_dense_fwd<KT, true><<<grid, COPY_BLOCK, lds, s>>>(              \
+          X, W, labels, wts, diff, loss_acc, inv_b, (int)B, (int)d);       \
+    else                                                                   \
+      k_lr_dense_fwd<KT, false><<<grid, COPY_BLOCK, lds, s>>>(             \
+          X, W, labels, wts, diff, loss_acc, inv_b, (int)B, (int)d);       \
     break
     LAUNCH_KT(1); LAUNCH_KT(2); LAUNCH_KT(3); LAUNCH_KT(4);
     LAUNCH_KT(5); LAUNCH_KT(6); LAUNCH_KT(7); LAUNCH_KT(8);

@@ -34,8 +34,10 @@ def bench(name, fn, iters=200):
     return dt
 
 
-bench("fused lr_dense_fwd", lambda: hip.lr_dense_fwd(
-    x, w, labels, None, diff, loss, 1.0 / B))
+bench("fused lr_dense_fwd nt", lambda: hip.lr_dense_fwd(
+    x, w, labels, None, diff, loss, 1.0 / B, True))
+bench("fused lr_dense_fwd plain", lambda: hip.lr_dense_fwd(
+    x, w, labels, None, diff, loss, 1.0 / B, False))
 logits = torch.empty(B, K, device=dev)
 
 
@@ -50,3 +52,11 @@ bench("post alone", lambda: hip.lr_dense_post(logits, labels, None, loss,
                                               1.0 / B))
 bench("bwd addmm (X^T@diff)", lambda: w.addmm_(x.t(), diff, alpha=-1e-6))
 print("peak-est: X bytes", B * d * 4 / 1e6, "MB")
+
+rows = torch.randint(0, 10_000_000, (1_000_000,), device=dev).unique()
+shard = torch.randn(10_000_000, 10, device=dev)
+gsq = torch.rand(10_000_000, 10, device=dev)
+vals = torch.randn(rows.numel(), 10, device=dev)
+bench("row_gather 1M rows K=10", lambda: hip.row_gather(shard, rows))
+bench("scatter_adagrad_u 1M rows", lambda: hip.row_scatter_adagrad(
+    shard, gsq, rows, vals, 0.05, 0.01, 1e-10, True))
