@@ -348,16 +348,19 @@ class AsyncEngine:
             for ww in w:
                 ww.wait()
             remotes.append(dst)
+        elem = torch.empty(0, dtype=table.dtype).element_size()
         for dst in remotes:
             cnt = torch.empty(1, dtype=torch.int64)
             dist.recv(cnt, dst, group=self.rep)
             k = int(cnt[0])
             if not k:
                 continue
-            ids = torch.empty(k, dtype=torch.int64)
-            dist.recv(ids, dst, group=self.rep)
-            vals = torch.empty(k * unit, dtype=table.dtype)
-            dist.recv(vals, dst, group=self.rep)
+            # count, then ONE [ids][vals] blob (matching the request
+            # wire format; ids first keeps the vals view 8-byte aligned)
+            blob = torch.empty(k * 8 + k * unit * elem, dtype=torch.uint8)
+            dist.recv(blob, dst, group=self.rep)
+            ids = blob[:k * 8].view(torch.int64)
+            vals = blob[k * 8:].view(table.dtype)
             cache[ids] = vals.view(k, unit).to(cache.device)
             total += k
         return total
@@ -514,8 +517,10 @@ class AsyncEngine:
             dist.send(torch.tensor([ids.numel()], dtype=torch.int64), src,
                       group=self.rep)
             if ids.numel():
-                dist.send(ids.cpu().contiguous(), src, group=self.rep)
-                dist.send(served.reshape(-1).cpu().contiguous(), src,
-                          group=self.rep)
+                blob = torch.cat(
+                    [ids.cpu().contiguous().view(torch.uint8).view(-1),
+                     served.reshape(-1).cpu().contiguous()
+                           .view(torch.uint8).view(-1)])
+                dist.send(blob, src, group=self.rep)
         else:
             CHECK(False, f"unknown async op {op}")
