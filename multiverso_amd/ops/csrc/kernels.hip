@@ -1255,7 +1255,14 @@ extern "C" void mv_launch_row_scatter_adagrad(
 // docs/ENGINEERING_NOTES.md carries the full argument.
 // ---------------------------------------------------------------------------
 
-template <int K>
+// KKC != 0 pins the runtime class count at compile time: the `k < KK`
+// guards in the unrolled row loop then constant-fold away.  With a
+// runtime KK the compiler emits a PER-CLASS BRANCH LADDER with a full
+// `s_waitcnt vmcnt(0)` after every gathered-row load — each feature's
+// row read serializes at full memory latency (the same disease measured
+// at 5x on k_lr_dense_fwd; see profiles/round2_dense_fused.md).  The
+// launcher dispatches exact K for K <= 16, guarded buckets beyond.
+template <int K, int KKC = 0>
 __global__ void k_lr_softmax_fwd(const float* __restrict__ w,
                                  const int64_t* __restrict__ keys,
                                  const float* __restrict__ vals,
@@ -1264,6 +1271,7 @@ __global__ void k_lr_softmax_fwd(const float* __restrict__ w,
                                  const float* __restrict__ wts,
                                  float* __restrict__ err,
                                  float* __restrict__ loss, int B, int KK) {
+  if (KKC != 0) KK = KKC;
   int wid = (int)((blockIdx.x * (int64_t)blockDim.x + threadIdx.x) >> 6);
   int lane = threadIdx.x & 63;
   int nwaves = (int)((gridDim.x * (int64_t)blockDim.x) >> 6);
@@ -1320,11 +1328,14 @@ __global__ void k_lr_softmax_scatter(float* __restrict__ w,
   int nwaves = (int)((gridDim.x * (int64_t)blockDim.x) >> 6);
   for (int i = wid; i < B; i += nwaves) {
     int jb = ptr[i], je = ptr[i + 1];
-    int64_t pairs = (int64_t)(je - jb) * K;
+    // u32 pair math: nnz*K < 2^32 always (B*nnz is int32 CSR already);
+    // a 64-bit div per pair was ~40 VALU cycles each
+    uint32_t pairs = (uint32_t)(je - jb) * (uint32_t)K;
     const float* e = err + (int64_t)i * K;
-    for (int64_t t = lane; t < pairs; t += 64) {
-      int j = jb + (int)(t / K);
-      int k = (int)(t % K);
+    for (uint32_t t = lane; t < pairs; t += 64) {
+      uint32_t q = t / (uint32_t)K;
+      int j = jb + (int)q;
+      int k = (int)(t - q * (uint32_t)K);
       int64_t idx = keys[j] * K + k;
       float g = vals[j] * e[k];
       if (reg_type == 1) {
@@ -1344,14 +1355,23 @@ extern "C" void mv_launch_lr_softmax_fwd(
     int64_t B, int64_t K, hipStream_t s) {
   if (!B) return;
   int grid = grid_for(B * 64);
-#define SMAX_CASE(KC)                                                        \
-  else if (K <= KC) k_lr_softmax_fwd<KC><<<grid, BLOCK, 0, s>>>(             \
-      w, keys, vals, ptr, labels, wts, err, loss, (int)B, (int)K)
-  if (false) {}
-  SMAX_CASE(2); SMAX_CASE(4); SMAX_CASE(8); SMAX_CASE(16);
-  SMAX_CASE(32); SMAX_CASE(64);
+#define SMAX_ARGS w, keys, vals, ptr, labels, wts, err, loss, (int)B, (int)K
+#define SMAX_EXACT(KC)                                                       \
+  case KC:                                                                   \
+    k_lr_softmax_fwd<KC, KC><<<grid, BLOCK, 0, s>>>(SMAX_ARGS);              \
+    return
+  switch (K) {
+    SMAX_EXACT(2); SMAX_EXACT(3); SMAX_EXACT(4); SMAX_EXACT(5);
+    SMAX_EXACT(6); SMAX_EXACT(7); SMAX_EXACT(8); SMAX_EXACT(9);
+    SMAX_EXACT(10); SMAX_EXACT(11); SMAX_EXACT(12); SMAX_EXACT(13);
+    SMAX_EXACT(14); SMAX_EXACT(15); SMAX_EXACT(16);
+    default: break;
+  }
+  if (K <= 32) k_lr_softmax_fwd<32><<<grid, BLOCK, 0, s>>>(SMAX_ARGS);
+  else if (K <= 64) k_lr_softmax_fwd<64><<<grid, BLOCK, 0, s>>>(SMAX_ARGS);
   else __builtin_trap();  // K > 64: host refuses before launch
-#undef SMAX_CASE
+#undef SMAX_EXACT
+#undef SMAX_ARGS
 }
 
 extern "C" void mv_launch_lr_softmax_scatter(
@@ -1388,7 +1408,7 @@ static __device__ __forceinline__ float ftrl_w(float z, float n,
   return (fabsf(z) > l1) ? w : 0.f;
 }
 
-template <int K>
+template <int K, int KKC = 0>
 __global__ void k_lr_ftrl_fwd(const float* __restrict__ zn,
                               const int64_t* __restrict__ keys,
                               const float* __restrict__ vals,
@@ -1399,6 +1419,7 @@ __global__ void k_lr_ftrl_fwd(const float* __restrict__ zn,
                               float* __restrict__ loss,
                               float alpha_inv, float beta, float l1,
                               float l2, int B, int KK) {
+  if (KKC != 0) KK = KKC;  // see k_lr_softmax_fwd: exact-K folds guards
   int wid = (int)((blockIdx.x * (int64_t)blockDim.x + threadIdx.x) >> 6);
   int lane = threadIdx.x & 63;
   int nwaves = (int)((gridDim.x * (int64_t)blockDim.x) >> 6);
@@ -1452,11 +1473,12 @@ __global__ void k_lr_ftrl_scatter(float* __restrict__ zn,
   int nwaves = (int)((gridDim.x * (int64_t)blockDim.x) >> 6);
   for (int i = wid; i < B; i += nwaves) {
     int jb = ptr[i], je = ptr[i + 1];
-    int64_t pairs = (int64_t)(je - jb) * K;
+    uint32_t pairs = (uint32_t)(je - jb) * (uint32_t)K;  // see softmax scatter
     const float* e = err + (int64_t)i * K;
-    for (int64_t t = lane; t < pairs; t += 64) {
-      int j = jb + (int)(t / K);
-      int k = (int)(t % K);
+    for (uint32_t t = lane; t < pairs; t += 64) {
+      uint32_t q = t / (uint32_t)K;
+      int j = jb + (int)q;
+      int k = (int)(t - q * (uint32_t)K);
       int64_t base = keys[j] * (2 * (int64_t)K);
       float z = zn[base + k];
       float n = zn[base + K + k];
@@ -1479,15 +1501,23 @@ extern "C" void mv_launch_lr_ftrl_fwd(
     hipStream_t s) {
   if (!B) return;
   int grid = grid_for(B * 64);
-#define FTRL_CASE(KC)                                                        \
-  else if (K <= KC) k_lr_ftrl_fwd<KC><<<grid, BLOCK, 0, s>>>(                \
-      zn, keys, vals, ptr, labels, wts, err, loss, alpha_inv, beta, l1,      \
-      l2, (int)B, (int)K)
-  if (false) {}
-  FTRL_CASE(1); FTRL_CASE(2); FTRL_CASE(4); FTRL_CASE(8); FTRL_CASE(16);
-  FTRL_CASE(32);
+#define FTRL_ARGS zn, keys, vals, ptr, labels, wts, err, loss, alpha_inv, \
+                  beta, l1, l2, (int)B, (int)K
+#define FTRL_EXACT(KC)                                                       \
+  case KC:                                                                   \
+    k_lr_ftrl_fwd<KC, KC><<<grid, BLOCK, 0, s>>>(FTRL_ARGS);                 \
+    return
+  switch (K) {
+    FTRL_EXACT(1); FTRL_EXACT(2); FTRL_EXACT(3); FTRL_EXACT(4);
+    FTRL_EXACT(5); FTRL_EXACT(6); FTRL_EXACT(7); FTRL_EXACT(8);
+    FTRL_EXACT(9); FTRL_EXACT(10); FTRL_EXACT(11); FTRL_EXACT(12);
+    FTRL_EXACT(13); FTRL_EXACT(14); FTRL_EXACT(15); FTRL_EXACT(16);
+    default: break;
+  }
+  if (K <= 32) k_lr_ftrl_fwd<32><<<grid, BLOCK, 0, s>>>(FTRL_ARGS);
   else __builtin_trap();  // K > 32: host refuses before launch
-#undef FTRL_CASE
+#undef FTRL_EXACT
+#undef FTRL_ARGS
 }
 
 extern "C" void mv_launch_lr_ftrl_scatter(
