@@ -762,10 +762,15 @@ __global__ void k_w2v(float* __restrict__ in_emb, float* __restrict__ out_emb,
     int ib = in_off[g], ie = in_off[g + 1];
     for (int i = ib; i < ie; ++i) {
       const float* row = in_emb + in_idx[i] * dim;
+      // clamped UNCONDITIONAL loads: a divergent `if (c < dim)` around
+      // the load compiled to a branch + s_waitcnt vmcnt(0) PER DPL
+      // GROUP (full-latency serial, the round-2 ladder disease).
+      // min() keeps the read in-bounds; inactive lanes accumulate a
+      // duplicate element that nothing reads (wv[d]=0 gates every use).
 #pragma unroll
       for (int d = 0; d < DPL; ++d) {
         int c = lane + 64 * d;
-        if (c < dim) h[d] += row[c];
+        h[d] += row[c < dim ? c : dim - 1];
       }
     }
     if (ie - ib > 1) {
@@ -778,10 +783,14 @@ __global__ void k_w2v(float* __restrict__ in_emb, float* __restrict__ out_emb,
       float* w = out_emb + out_idx[o] * dim;
       float wv[DPL];
       float f = 0.f;
+      // unconditional clamped load + select (see the h loop above):
+      // wv[d] MUST be 0 for inactive lanes (it feeds the wave-reduced
+      // dot), so the select stays — but on the value, not the load
 #pragma unroll
       for (int d = 0; d < DPL; ++d) {
         int c = lane + 64 * d;
-        wv[d] = (c < dim) ? w[c] : 0.f;
+        float x = w[c < dim ? c : dim - 1];
+        wv[d] = (c < dim) ? x : 0.f;
         f += h[d] * wv[d];
       }
 #pragma unroll
@@ -811,6 +820,22 @@ __global__ void k_w2v(float* __restrict__ in_emb, float* __restrict__ out_emb,
     }
     for (int i = ib; i < ie; ++i) {
       float* row = in_emb + in_idx[i] * dim;
+      if (!ADAGRAD && !ATOMIC) {
+        // hogwild-default path: branch-free rmw — unconditional clamped
+        // loads (issue back-to-back under one wait), guarded stores
+        float cur[DPL];
+#pragma unroll
+        for (int d = 0; d < DPL; ++d) {
+          int c = lane + 64 * d;
+          cur[d] = row[c < dim ? c : dim - 1];
+        }
+#pragma unroll
+        for (int d = 0; d < DPL; ++d) {
+          int c = lane + 64 * d;
+          if (c < dim) row[c] = cur[d] + lr * err[d];
+        }
+        continue;
+      }
 #pragma unroll
       for (int d = 0; d < DPL; ++d) {
         int c = lane + 64 * d;
@@ -862,10 +887,15 @@ __global__ void k_w2v_ns(float* __restrict__ in_emb, float* __restrict__ out_emb
     int ib = in_off ? in_off[g] : g, ie = in_off ? in_off[g + 1] : g + 1;
     for (int i = ib; i < ie; ++i) {
       const float* row = in_emb + in_idx[i] * dim;
+      // clamped UNCONDITIONAL loads: a divergent `if (c < dim)` around
+      // the load compiled to a branch + s_waitcnt vmcnt(0) PER DPL
+      // GROUP (full-latency serial, the round-2 ladder disease).
+      // min() keeps the read in-bounds; inactive lanes accumulate a
+      // duplicate element that nothing reads (wv[d]=0 gates every use).
 #pragma unroll
       for (int d = 0; d < DPL; ++d) {
         int c = lane + 64 * d;
-        if (c < dim) h[d] += row[c];
+        h[d] += row[c < dim ? c : dim - 1];
       }
     }
     if (ie - ib > 1) {
@@ -890,10 +920,14 @@ __global__ void k_w2v_ns(float* __restrict__ in_emb, float* __restrict__ out_emb
       float* w = out_emb + node * dim;
       float wv[DPL];
       float f = 0.f;
+      // unconditional clamped load + select (see the h loop above):
+      // wv[d] MUST be 0 for inactive lanes (it feeds the wave-reduced
+      // dot), so the select stays — but on the value, not the load
 #pragma unroll
       for (int d = 0; d < DPL; ++d) {
         int c = lane + 64 * d;
-        wv[d] = (c < dim) ? w[c] : 0.f;
+        float x = w[c < dim ? c : dim - 1];
+        wv[d] = (c < dim) ? x : 0.f;
         f += h[d] * wv[d];
       }
 #pragma unroll
@@ -923,6 +957,22 @@ __global__ void k_w2v_ns(float* __restrict__ in_emb, float* __restrict__ out_emb
     }
     for (int i = ib; i < ie; ++i) {
       float* row = in_emb + in_idx[i] * dim;
+      if (!ADAGRAD && !ATOMIC) {
+        // hogwild-default path: branch-free rmw — unconditional clamped
+        // loads (issue back-to-back under one wait), guarded stores
+        float cur[DPL];
+#pragma unroll
+        for (int d = 0; d < DPL; ++d) {
+          int c = lane + 64 * d;
+          cur[d] = row[c < dim ? c : dim - 1];
+        }
+#pragma unroll
+        for (int d = 0; d < DPL; ++d) {
+          int c = lane + 64 * d;
+          if (c < dim) row[c] = cur[d] + lr * err[d];
+        }
+        continue;
+      }
 #pragma unroll
       for (int d = 0; d < DPL; ++d) {
         int c = lane + 64 * d;
