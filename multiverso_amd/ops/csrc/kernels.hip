@@ -445,6 +445,18 @@ __global__ void k_row_gather_f4(float4* __restrict__ out,
   }
 }
 
+__global__ void k_row_gather_f4_u32(float4* __restrict__ out,
+                                    const float4* __restrict__ shard,
+                                    const int64_t* __restrict__ rows,
+                                    uint32_t total4, uint32_t cols4) {
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total4;
+       i += stride) {
+    uint32_t r = i / cols4, c = i - r * cols4;
+    out[i] = shard[rows[r] * (int64_t)cols4 + c];
+  }
+}
+
 __global__ void k_row_gather(float* __restrict__ out,
                              const float* __restrict__ shard,
                              const int64_t* __restrict__ rows,
@@ -681,7 +693,11 @@ void mv_launch_row_gather(float* out, const float* shard, const int64_t* rows,
                           int64_t nrows, int64_t cols, hipStream_t s) {
   if (!nrows || !cols) return;
   int64_t total = nrows * cols;
-  if (cols % 4 == 0) {
+  if (cols % 4 == 0 && total / 4 <= UINT32_MAX) {
+    k_row_gather_f4_u32<<<grid_for(total / 4), BLOCK, 0, s>>>(
+        (float4*)out, (const float4*)shard, rows, (uint32_t)(total / 4),
+        (uint32_t)(cols / 4));
+  } else if (cols % 4 == 0) {
     int64_t cols4 = cols / 4;
     k_row_gather_f4<<<grid_for(nrows * cols4), BLOCK, 0, s>>>(
         (float4*)out, (const float4*)shard, rows, nrows, cols4);
