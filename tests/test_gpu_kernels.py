@@ -89,7 +89,7 @@ def test_dcasgda(hip, n):
     assert torch.equal(bak, data)
 
 
-@pytest.mark.parametrize("cols", [128, 200, 7])
+@pytest.mark.parametrize("cols", [128, 200, 10, 2, 7])
 def test_row_gather(hip, cols):
     shard = rand(500 * cols, 11).view(500, cols)
     rows = torch.randint(0, 500, (64,), dtype=torch.int64).cuda()
@@ -99,7 +99,7 @@ def test_row_gather(hip, cols):
     assert torch.equal(out, ref)
 
 
-@pytest.mark.parametrize("cols", [128, 7])
+@pytest.mark.parametrize("cols", [128, 10, 7])
 def test_row_scatter_add(hip, cols):
     shard = rand(300 * cols, 12).view(300, cols)
     ref = shard.clone()
