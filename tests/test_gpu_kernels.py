@@ -826,3 +826,34 @@ def test_lr_dense_fwd_lds_fallback(hip):
     diff = torch.empty(B, K, device="cuda:0")
     loss_acc = torch.zeros((), device="cuda:0")
     assert not hip.lr_dense_fwd(x, w, labels, None, diff, loss_acc, 1.0 / B)
+
+
+@pytest.mark.parametrize("cols,assume_unique", [(10, True), (10, False),
+                                                (128, True), (7, False)])
+def test_row_scatter_adagrad_vs_torch(hip, cols, assume_unique):
+    """Keyed AdaGrad scatter (K15+K4, incl. the u32 loop-index forms)
+    vs the torch formula — unique row ids so both the atomic and plain
+    variants are deterministic."""
+    torch.manual_seed(3 + cols)
+    R, n = 5000, 700
+    shard = (torch.randn(R, cols) * 0.1).cuda()
+    gsq = torch.rand(R, cols).cuda()
+    rows = torch.randperm(R)[:n].cuda()
+    vals = torch.randn(n, cols).cuda()
+    lr, rho, eps = 0.05, 0.01, 1e-10
+
+    ref_shard, ref_gsq = shard.clone(), gsq.clone()
+    g = vals / lr
+    Gn = ref_gsq[rows] + g * g
+    ref_gsq[rows] = Gn
+    step = rho * g / torch.sqrt(Gn + eps)
+    step = torch.where(g == 0, torch.zeros_like(step), step)
+    ref_shard[rows] -= step
+
+    hip.row_scatter_adagrad(shard, gsq, rows, vals.contiguous(), lr, rho,
+                            eps, assume_unique)
+    torch.cuda.synchronize()
+    assert torch.allclose(gsq, ref_gsq, rtol=1e-5, atol=1e-6), \
+        (gsq - ref_gsq).abs().max()
+    assert torch.allclose(shard, ref_shard, rtol=1e-4, atol=1e-6), \
+        (shard - ref_shard).abs().max()
