@@ -567,6 +567,7 @@ bool lr_dense_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor labels,
              nt ? 1 : 0, cur_stream()) != 0;
 }
 
+
 void lr_ftrl_scatter(torch::Tensor zn, torch::Tensor keys,
                      torch::Tensor vals, torch::Tensor ptr,
                      torch::Tensor err, double alpha, double beta, double l1,

@@ -1888,3 +1888,4 @@ extern "C" int mv_launch_lr_dense_fwd(const float* X, const float* W,
   }
   return 1;
 }
+

@@ -857,3 +857,5 @@ def test_row_scatter_adagrad_vs_torch(hip, cols, assume_unique):
         (gsq - ref_gsq).abs().max()
     assert torch.allclose(shard, ref_shard, rtol=1e-4, atol=1e-6), \
         (shard - ref_shard).abs().max()
+
+
