@@ -134,7 +134,11 @@ class PSModel:
         if batches:
             cat_keys = torch.cat([b.keys for b in batches])
             if self.cfg.input_size <= (1 << 31) - 1:
-                # keys fit int32: the dedup sort runs at half the bytes
+                # keys fit int32: the dedup sort runs at half the bytes.
+                # (An explicit torch.sort + manual inverse/masked-select
+                # was measured 10-15% SLOWER whole-app than unique —
+                # the extra elementwise/scatter kernels outweigh any
+                # sort-path difference; keep unique.)
                 union, inv = torch.unique(cat_keys.to(torch.int32),
                                           return_inverse=True)
                 union = union.to(torch.int64)
