@@ -68,6 +68,12 @@ def test_matrix_rows_duplicate(env):
     t = mv.MatrixTable(10, 2)
     t.add_rows([3, 3], torch.ones(2, 2))
     assert torch.equal(t.get_rows([3]), torch.full((1, 2), 2.0))
+    # duplicate ids in a GET return one row per occurrence, caller order
+    # (exercises the ws=1 direct-gather fast path)
+    got = t.get_rows([3, 0, 3])
+    assert torch.equal(got, torch.stack([torch.full((2,), 2.0),
+                                         torch.zeros(2),
+                                         torch.full((2,), 2.0)]))
 
 
 def test_matrix_random_init(env):
