@@ -9,6 +9,7 @@ synthetic data / random-init weights)."""
 
 from __future__ import annotations
 
+from itertools import repeat
 from typing import Iterator, List, Optional, Tuple
 
 import torch
@@ -40,18 +41,27 @@ class TextBlockReader:
         self.world = world
 
     def blocks(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:
+        # Hot loop: one python-level dict probe per TOKEN dominates the
+        # end-to-end CLI rate (the GPU trains a 500k-word block in
+        # ~7.6 ms; a naive comprehension with method-attribute lookups
+        # parsed at ~6M words/s). Bound-method map() + local names is
+        # the fastest pure-python form measured.
         words: List[int] = []
         sids: List[int] = []
         sid = 0
         taken = 0
+        get = self.dict._index.get
+        m1 = repeat(-1)
+        world, rank = self.world, self.rank
         with open(self.path) as f:
             for snum, line in enumerate(f):
-                if snum % self.world != self.rank:
+                if snum % world != rank:
                     continue
-                ids = [self.dict.get_id(w) for w in line.split()]
-                ids = [i for i in ids if i >= 0][:MAX_SENTENCE_LEN]
+                ids = [i for i in map(get, line.split(), m1) if i >= 0]
                 if not ids:
                     continue
+                if len(ids) > MAX_SENTENCE_LEN:
+                    ids = ids[:MAX_SENTENCE_LEN]
                 words.extend(ids)
                 sids.extend([sid] * len(ids))
                 sid += 1
